@@ -1,0 +1,116 @@
+"""Greedy pathfinder pinned against the reference's expected SSA paths and
+costs (contractionpath/paths/cotengrust.rs:229-307)."""
+
+from tnc_amd import CompositeTensor, Greedy, LeafTensor, RandomGreedy
+from tnc_amd.contraction_path import path, validate_path
+from tnc_amd.paths import BasicContractionPathResult
+
+
+def _setup_simple():
+    bd = {0: 5, 1: 2, 2: 6, 3: 8, 4: 1, 5: 3, 6: 4}
+    return CompositeTensor(
+        [
+            LeafTensor.new_from_map([4, 3, 2], bd),
+            LeafTensor.new_from_map([0, 1, 3, 2], bd),
+            LeafTensor.new_from_map([4, 5, 6], bd),
+        ]
+    )
+
+
+def _setup_complex():
+    bd = {0: 27, 1: 18, 2: 12, 3: 15, 4: 5, 5: 3, 6: 18, 7: 22, 8: 45, 9: 65,
+          10: 5, 11: 17}
+    return CompositeTensor(
+        [
+            LeafTensor.new_from_map([4, 3, 2], bd),
+            LeafTensor.new_from_map([0, 1, 3, 2], bd),
+            LeafTensor.new_from_map([4, 5, 6], bd),
+            LeafTensor.new_from_map([6, 8, 9], bd),
+            LeafTensor.new_from_map([10, 8, 9], bd),
+            LeafTensor.new_from_map([5, 1, 0], bd),
+        ]
+    )
+
+
+def test_greedy_simple():
+    """cotengrust.rs:229-243."""
+    result = Greedy().find_path(_setup_simple())
+    assert result == BasicContractionPathResult(path((0, 1), (3, 2)), 600.0, 538.0)
+
+
+def test_greedy_simple_inner():
+    """cotengrust.rs:246-259."""
+    bd = {0: 5, 1: 2, 2: 6, 3: 8, 4: 1, 5: 3, 6: 4}
+    tn = CompositeTensor(
+        [
+            LeafTensor.new_from_map([4, 3, 2], bd),
+            LeafTensor.new_from_map([4, 3, 2], bd),
+            LeafTensor.new_from_map([0, 1, 5], bd),
+            LeafTensor.new_from_map([1, 6], bd),
+        ]
+    )
+    result = Greedy().find_path(tn)
+    assert result == BasicContractionPathResult(
+        path((0, 1), (2, 3), (4, 5)), 228.0, 121.0
+    )
+
+
+def test_greedy_simple_outer():
+    """cotengrust.rs:262-275."""
+    bd = {0: 3, 1: 2, 2: 2}
+    tn = CompositeTensor(
+        [
+            LeafTensor.new_from_map([0], bd),
+            LeafTensor.new_from_map([1], bd),
+            LeafTensor.new_from_map([2], bd),
+        ]
+    )
+    result = Greedy().find_path(tn)
+    assert result == BasicContractionPathResult(path((2, 1), (0, 3)), 16.0, 19.0)
+
+
+def test_greedy_complex_outer():
+    """cotengrust.rs:278-291."""
+    bd = {0: 5, 1: 4}
+    tn = CompositeTensor(
+        [
+            LeafTensor.new_from_map([0], bd),
+            LeafTensor.new_from_map([0], bd),
+            LeafTensor.new_from_map([1], bd),
+            LeafTensor.new_from_map([1], bd),
+        ]
+    )
+    result = Greedy().find_path(tn)
+    assert result == BasicContractionPathResult(
+        path((0, 1), (2, 3), (5, 4)), 10.0, 11.0
+    )
+
+
+def test_greedy_complex():
+    """cotengrust.rs:294-307."""
+    result = Greedy().find_path(_setup_complex())
+    assert result == BasicContractionPathResult(
+        path((1, 5), (3, 4), (6, 0), (7, 2), (9, 8)), 529815.0, 89478.0
+    )
+
+
+def test_random_greedy_not_worse():
+    tn = _setup_complex()
+    g = Greedy().find_path(tn)
+    rg = RandomGreedy(10).find_path(tn)
+    assert rg.flops <= g.flops
+    validate_path(rg.replace_path())
+
+
+def test_greedy_nested_composites():
+    """find_path recurses into composites (cotengrust.rs:120-137)."""
+    bd = {0: 2, 1: 2, 2: 2, 3: 2, 4: 2}
+    inner = CompositeTensor(
+        [LeafTensor.new_from_map([0, 1], bd), LeafTensor.new_from_map([1, 2], bd)]
+    )
+    tn = CompositeTensor([inner, LeafTensor.new_from_map([2, 3], bd),
+                          LeafTensor.new_from_map([3, 4], bd)])
+    result = Greedy().find_path(tn)
+    assert 0 in result.ssa_path.nested
+    replace = result.replace_path()
+    validate_path(replace)
