@@ -1,0 +1,344 @@
+#!/usr/bin/env python3
+"""Benchmark: pairwise-contraction GFLOP/s (c128) on the frozen 36-qubit RQC
+amplitude network (BASELINE.json configs[2]; --gpus N>1 = configs[3],
+one partition per GPU over RCCL/xGMI).
+
+Protocol mirrors the reference benchmark (benchmark/src/main.rs:355-405):
+the contraction plan (path / partitioning) is precomputed (frozen in the
+fixture / derived deterministically) and excluded from timing; the clock
+runs from the pre-step barrier to the final tensor on GPU 0. A "step" is one
+full contraction of the fixture network. value = executed metric flops
+(sum over path steps of (8*s-2)*o, contraction_cost.rs:26-32) / wall time,
+aggregated over all ranks.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--fixture rqc36]
+Multi-GPU: launched via torch.distributed.run, one rank per GPU (RCCL).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+# MI355X peaks (MI355X_MICROARCH.md; f64 matrix peak cross-checked by the
+# committed muBench under profiles/ — see DESIGN.md "Measurement")
+F64_MFMA_PEAK = 78.6e12   # real flops/s, v_mfma_f64_16x16x4_f64 dense
+HBM_PEAK = 8.0e12         # bytes/s (spec; ~6.3 TB/s achievable)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--fixture", default="rqc36")
+    p.add_argument("--no-cpu-baseline", action="store_true")
+    p.add_argument("--trials", type=int, default=16,
+                   help="random-greedy trials for the multi-GPU plan")
+    return p.parse_args()
+
+
+def step_bytes(info):
+    """Algorithmic bytes of one einsum step: read A and B once, write out
+    once (16 B per c128 element)."""
+    return 16.0 * (info.m * info.k + info.k * info.n + info.m * info.n)
+
+
+def roofline_from_profile(infos, step_ms, gemm_ms, kinds):
+    """Dominant kernel + its roofline leg from live HIP-event timings."""
+    dom = max(range(len(step_ms)), key=lambda s: step_ms[s])
+    info = infos[dom]
+    traffic = None
+    pmc_path = os.path.join(ROOT, "profiles", "pmc_calibration.json")
+    if os.path.exists(pmc_path):
+        try:
+            with open(pmc_path) as f:
+                pmc = json.load(f)
+            traffic = pmc.get("dominant_kernel_traffic_bytes")
+        except Exception:
+            traffic = None
+    if kinds[dom] >= 2 and gemm_ms[dom] > 0:
+        dur_s = gemm_ms[dom] / 1e3
+        achieved = info.flops / dur_s
+        return {
+            "bound": "mfma",
+            "achieved": achieved,
+            "peak": F64_MFMA_PEAK,
+            "unit": "FLOP/s",
+            "frac": achieved / F64_MFMA_PEAK,
+            "traffic": traffic,
+            "kernel": "k_zgemm_mfma",
+            "launch_ms": gemm_ms[dom],
+            "mnk": [info.m, info.n, info.k],
+        }
+    dur_s = step_ms[dom] / 1e3
+    achieved = step_bytes(info) / dur_s
+    return {
+        "bound": "hbm",
+        "achieved": achieved,
+        "peak": HBM_PEAK,
+        "unit": "B/s",
+        "frac": achieved / HBM_PEAK,
+        "traffic": traffic,
+        "kernel": "k_einsum_smallk",
+        "launch_ms": step_ms[dom],
+        "mnk": [info.m, info.n, info.k],
+    }
+
+
+def cpu_baseline(infos, budget_s=15.0, cap_elems=2 ** 27):
+    """Oracle (numpy einsum -> BLAS zgemm) timed on the host cores over a
+    bounded sample: the largest path steps whose operands fit `cap_elems`,
+    random-valued inputs of the same shapes (einsum time is value-
+    independent), until ~budget_s of wall. kind="port"."""
+    import numpy as np
+
+    import oracle
+
+    order = sorted(range(len(infos)), key=lambda s: -infos[s].flops)
+    rng = np.random.default_rng(0)
+    total_flops = 0.0
+    total_time = 0.0
+    used = 0
+    for s in order:
+        info = infos[s]
+        elems = info.m * info.k + info.k * info.n + info.m * info.n
+        if elems > cap_elems:
+            continue
+        # synthetic operands shaped like the step ([M legs][K legs] etc. as
+        # flat 2-leg tensors: zgemm time depends on (M, N, K) only)
+        a = (rng.standard_normal((int(info.m), int(info.k)))
+             + 1j * rng.standard_normal((int(info.m), int(info.k))))
+        b = (rng.standard_normal((int(info.k), int(info.n)))
+             + 1j * rng.standard_normal((int(info.k), int(info.n))))
+        t0 = time.perf_counter()
+        oracle.contract_ndarrays([0, 2], [0, 1], a, [1, 2], b)
+        dt = time.perf_counter() - t0
+        total_time += dt
+        total_flops += info.flops
+        used += 1
+        if total_time >= budget_s:
+            break
+    if total_time == 0.0:
+        return None
+    try:
+        cores = len(os.sched_getaffinity(0))
+    except AttributeError:
+        cores = os.cpu_count()
+    return {
+        "value": total_flops / total_time / 1e9,
+        "unit": "GFLOP/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{used} largest rqc36 path steps with operands <= "
+                  f"{cap_elems} elems, random-valued, {total_time:.1f}s",
+    }
+
+
+def emit(result):
+    print(json.dumps(result), flush=True)
+
+
+def run_single(args):
+    from tnc_amd.contraction_path import ContractionPath
+    from tnc_amd.executor import ContractionEngine
+    from tnc_amd.fixtures import load_fixture
+
+    tn, replace_toplevel, meta = load_fixture(args.fixture)
+    replace = ContractionPath.simple(replace_toplevel)
+    eng = ContractionEngine(tn, replace, device=0)
+    flops_per_contraction = eng.total_flops
+
+    # one profiled pass (doubles as extra warmup)
+    _, step_ms, gemm_ms, kinds = eng.contract_profiled()
+    roofline = roofline_from_profile(eng.infos, step_ms, gemm_ms, kinds)
+
+    for _ in range(args.warmup):
+        eng.contract()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        eng.contract()
+    wall = time.perf_counter() - t0
+
+    value = flops_per_contraction * args.steps / wall / 1e9
+    cb = None if args.no_cpu_baseline else cpu_baseline(eng.infos)
+    eng.close()
+    emit({
+        "metric": "pairwise-contraction GFLOP/s (c128)",
+        "value": value,
+        "unit": "GFLOP/s",
+        "n_gpus": 1,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": wall / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "c128",
+        "data": "synthetic",
+        "config": {
+            "workload": f"{args.fixture}: 36q depth-14 RQC single-amplitude "
+                        "network, frozen random-greedy path"
+                        if args.fixture == "rqc36" else args.fixture,
+            "tensors": len(tn.tensors),
+            "path_steps": len(replace_toplevel),
+            "metric_flops_per_contraction": flops_per_contraction,
+        },
+        "roofline": roofline,
+        "cpu_baseline": cb,
+    })
+
+
+def run_distributed(args):
+    import torch
+    import torch.distributed as dist_t
+
+    from tnc_amd import hiplib
+    from tnc_amd.contraction_path import ContractionPath
+    from tnc_amd.dist import make_plan, run_fanin
+    from tnc_amd.executor import ContractionEngine
+    from tnc_amd.fixtures import load_fixture
+    from tnc_amd.tensor import CompositeTensor
+
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    torch.cuda.set_device(local_rank)
+    dist_t.init_process_group("nccl")
+    device = torch.device(f"cuda:{local_rank}")
+
+    tn, _, meta = load_fixture(args.fixture)
+    plan = make_plan(tn, world, trials=args.trials, size_cap=4.0e9)
+    my_part = None
+    for p, r in plan.part_rank.items():
+        if r == rank:
+            my_part = p
+    eng = None
+    if my_part is not None:
+        sub = plan.partitioned.tensors[my_part]
+        if isinstance(sub, CompositeTensor):
+            inner = plan.path.nested.get(my_part, ContractionPath.simple([]))
+            eng = ContractionEngine(sub, inner, device=local_rank)
+        else:
+            eng = ContractionEngine(CompositeTensor([sub]),
+                                    ContractionPath.simple([]),
+                                    device=local_rank)
+
+    L = hiplib.lib()
+    import math as _math
+
+    class GpuBackend:
+        def __init__(self):
+            self.keep = []
+            self.nets = []
+
+        def ptr_of(self, handle):
+            kind, obj = handle
+            if kind == "net":
+                return L.tn_net_result_dev(obj)
+            return obj.data_ptr()
+
+        def send(self, handle, legs, dims, peer):
+            elems = int(_math.prod(dims)) if dims else 1
+            t = torch.empty((elems, 2), dtype=torch.float64, device=device)
+            hiplib.check(L.tn_memcpy_dtod(t.data_ptr(), self.ptr_of(handle),
+                                          elems * 16), "tn_memcpy_dtod")
+            dist_t.send(t, dst=peer)
+
+        def recv(self, legs, dims, peer):
+            elems = int(_math.prod(dims)) if dims else 1
+            t = torch.empty((elems, 2), dtype=torch.float64, device=device)
+            dist_t.recv(t, src=peer)
+            self.keep.append(t)
+            return ("torch", t)
+
+        def contract_pair(self, a, a_legs, a_dims, b, b_legs, b_dims):
+            net = L.tn_net_create(local_rank)
+            if not net:
+                raise RuntimeError(hiplib.last_error())
+            self.nets.append(net)
+            ia = L.tn_net_add_leaf_dev(net, hiplib._u64arr(a_legs),
+                                       hiplib._u64arr(a_dims), len(a_legs),
+                                       self.ptr_of(a))
+            ib = L.tn_net_add_leaf_dev(net, hiplib._u64arr(b_legs),
+                                       hiplib._u64arr(b_dims), len(b_legs),
+                                       self.ptr_of(b))
+            assert ia == 0 and ib == 1
+            pairs = hiplib._u64arr([0, 1])
+            hiplib.check(L.tn_net_contract(net, pairs, 1, None),
+                         "tn_net_contract(pair)")
+            return ("net", net)
+
+        def cleanup(self):
+            for net in self.nets:
+                L.tn_net_destroy(net)
+            self.nets.clear()
+            self.keep.clear()
+
+    times = []
+    final_host = None
+    for it in range(args.warmup + args.steps):
+        backend = GpuBackend()
+        dist_t.barrier()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        local = None
+        if eng is not None:
+            eng.contract()
+            local = ("net", eng.net)
+        final = run_fanin(plan, rank, local, backend.send, backend.recv,
+                          backend.contract_pair)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        # max over ranks
+        dt_t = torch.tensor([dt], dtype=torch.float64, device=device)
+        dist_t.all_reduce(dt_t, op=dist_t.ReduceOp.MAX)
+        if it >= args.warmup:
+            times.append(dt_t.item())
+        backend.cleanup()
+
+    if rank == 0:
+        wall = sum(times)
+        flops = plan.total_flops()
+        value = flops * args.steps / wall / 1e9
+        emit({
+            "metric": "pairwise-contraction GFLOP/s (c128)",
+            "value": value,
+            "unit": "GFLOP/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": wall / args.steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "c128",
+            "data": "synthetic",
+            "config": {
+                "workload": f"{args.fixture}: 36q depth-14 RQC amplitude, "
+                            f"{plan.nparts}-way partition, RCCL fan-in",
+                "tensors": len(tn.tensors),
+                "partitions": plan.nparts,
+                "metric_flops_per_contraction": flops,
+            },
+            "roofline": None,
+            "cpu_baseline": None,
+        })
+    dist_t.destroy_process_group()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        run_distributed(args)
+    else:
+        run_single(args)
+
+
+if __name__ == "__main__":
+    main()

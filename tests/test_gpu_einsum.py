@@ -1,0 +1,148 @@
+"""GPU einsum parity vs the oracle through the C ABI (tn_einsum_c128).
+
+Covers every kernel class: smallk (gate-apply shapes), anyk (skinny),
+dot (scalar out), outer product (K=1), MFMA GEMM, ragged GEMM (v1), packed
+and pack-free operands, unpack (caller-chosen out order), strided views,
+pow2 and non-pow2 dims.
+"""
+
+import numpy as np
+import pytest
+
+import oracle
+from oracle import OTensor
+from oracle.core import symmetric_difference
+
+pytestmark = pytest.mark.gpu
+
+
+def _rand(shape, rng):
+    return (rng.standard_normal(shape) + 1j * rng.standard_normal(shape)).astype(
+        np.complex128
+    )
+
+
+def run_case(a_labels, a_shape, b_labels, b_shape, seed=0, out_labels=None,
+             rtol=1e-12):
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(seed)
+    a = _rand(a_shape, rng)
+    b = _rand(b_shape, rng)
+    if out_labels is None:
+        dims_a, dims_b = list(a.shape), list(b.shape)
+        out_labels, _ = symmetric_difference(a_labels, dims_a, b_labels, dims_b)
+    ref = oracle.contract_ndarrays(out_labels, a_labels, a, b_labels, b)
+    got = hiplib.einsum_c128(out_labels, a_labels, a, b_labels, b)
+    np.testing.assert_allclose(got, ref, rtol=rtol, atol=1e-10)
+    return got
+
+
+def test_golden_pairwise():
+    """The reference's own golden vectors through the GPU path."""
+    import os
+
+    from tnc_amd import hiplib
+
+    z = np.load(os.path.join(os.path.dirname(__file__), "golden",
+                             "contraction_ref.npz"))
+    B, A = z["B_data"], z["A_data"]
+    bl, al = z["B_legs"].tolist(), z["A_legs"].tolist()
+    out_labels, _ = symmetric_difference(bl, B.shape, al, A.shape)
+    got = hiplib.einsum_c128(out_labels, bl, B, al, A)
+    assert out_labels == z["AxB_legs"].tolist()
+    np.testing.assert_allclose(got, z["AxB_data"], atol=1e-14)
+
+    C = z["C_data"]
+    cl = z["C_legs"].tolist()
+    out_labels, _ = symmetric_difference(cl, C.shape, bl, B.shape)
+    got = hiplib.einsum_c128(out_labels, cl, C, bl, B)
+    assert out_labels == z["BxC_legs"].tolist()
+    np.testing.assert_allclose(got, z["BxC_data"], atol=1e-14)
+
+
+def test_smallk_pow2():
+    # gate application: A = big state (rank 8), B = 2q gate
+    run_case([0, 1, 2, 3, 4, 5, 6, 7], [2] * 8, [10, 11, 2, 5], [2] * 4)
+
+
+def test_smallk_nonpow2():
+    run_case([0, 1, 2], [3, 5, 7], [2, 3], [7, 4])
+
+
+def test_outer_product():
+    run_case([0, 1], [4, 5], [2], [6])
+    run_case([0], [3], [1], [2])
+
+
+def test_scalar_dot_small():
+    run_case([0, 1], [2, 2], [0, 1], [2, 2])
+
+
+def test_scalar_dot_large():
+    # K = 2^18 -> dot kernel
+    run_case(list(range(18)), [2] * 18, list(range(18)), [2] * 18)
+
+
+def test_rank0_operand():
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(3)
+    a = _rand((), rng)
+    b = _rand((2, 3), rng)
+    ref = oracle.contract_ndarrays([7, 8], [], a, [7, 8], b)
+    got = hiplib.einsum_c128([7, 8], [], a, [7, 8], b)
+    np.testing.assert_allclose(got, ref, rtol=1e-12)
+
+
+def test_gemm_mfma_pow2():
+    # M = N = K = 256, pack-free A ([M..,K..] already) and packed B
+    run_case([0, 1, 2, 3], [16, 16, 16, 16], [4, 5, 2, 3], [16, 16, 16, 16])
+
+
+def test_gemm_mfma_qubit_legs():
+    # all dims 2: M=2^9, N=2^8, K=2^9
+    a_labels = list(range(18))          # 9 M legs + 9 K legs
+    b_labels = list(range(9, 18)) + list(range(100, 108))  # 9 K + 8 N
+    run_case(a_labels, [2] * 18, b_labels, [2] * 17, rtol=1e-11)
+
+
+def test_gemm_ragged_v1():
+    # non-pow2 dims -> v1 kernel with bounds handling
+    run_case([0, 1], [100, 9], [1, 2], [9, 75])
+    run_case([0, 1], [67, 130], [1, 2], [130, 41])
+
+
+def test_gemm_needs_packing_both():
+    # A legs ordered K-first, B legs N-first -> both packs run
+    run_case([2, 3, 0, 1], [8, 8, 16, 16], [5, 2, 3, 4], [16, 8, 8, 16])
+
+
+def test_gemm_unpack_interleaved_out():
+    # caller requests interleaved out order -> unpack permute
+    run_case([0, 1, 2], [32, 8, 64], [2, 3, 4], [64, 8, 32],
+             out_labels=[0, 3, 1, 4])
+
+
+def test_strided_views():
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(11)
+    a = _rand((8, 12, 6), rng).transpose(2, 0, 1)  # non-contiguous view
+    b = _rand((12, 10), rng)[::1, :]
+    ref = oracle.contract_ndarrays([0, 1], [2, 0, 3], a, [3, 1], b)
+    got = hiplib.einsum_c128([0, 1], [2, 0, 3], a, [3, 1], b)
+    np.testing.assert_allclose(got, ref, rtol=1e-12)
+
+
+def test_skinny_anyk():
+    # M=4, N=2^12, K=2^10 -> anyk kernel (skinny, K>64)
+    run_case([0, 1] + list(range(10, 20)), [2, 2] + [2] * 10,
+             list(range(10, 20)) + list(range(30, 42)), [2] * 22)
+
+
+def test_high_rank():
+    # rank-24 output, dims 2
+    a_labels = list(range(16))
+    b_labels = list(range(8, 24))
+    run_case(a_labels, [2] * 16, b_labels, [2] * 16, rtol=1e-11)

@@ -1,0 +1,140 @@
+"""Full-network contraction on the GPU (tn_net executor) vs the oracle and
+the reference's exact values."""
+
+import math
+
+import numpy as np
+import pytest
+
+from oracle import contract_network
+from oracle.adapters import network_to_otensors
+
+pytestmark = pytest.mark.gpu
+
+S2 = 1 / math.sqrt(2)
+
+
+def gpu_contract(tn, pathfinder=None):
+    from tnc_amd import Greedy
+    from tnc_amd.executor import contract_tensor_network_gpu
+
+    pathfinder = pathfinder or Greedy()
+    replace = pathfinder.find_path(tn).replace_path()
+    legs, data = contract_tensor_network_gpu(tn, replace)
+    ref = contract_network(network_to_otensors(tn), replace)
+    return legs, data, ref
+
+
+def test_ghz_statevector():
+    from tnc_amd import Circuit, TensorData
+
+    c = Circuit()
+    qr = c.allocate_register(3)
+    c.append_gate(TensorData.from_gate("h"), [qr.qubit(0)])
+    c.append_gate(TensorData.from_gate("cx"), [qr.qubit(0), qr.qubit(1)])
+    c.append_gate(TensorData.from_gate("cx"), [qr.qubit(1), qr.qubit(2)])
+    tn, permutor = c.into_statevector_network()
+    legs, data, ref = gpu_contract(tn)
+    assert legs == ref.legs
+    np.testing.assert_allclose(data, ref.data, atol=1e-15)
+    _, _, out = permutor.apply(legs, list(data.shape), data)
+    sv = out.reshape(-1)
+    expect = np.zeros(8, dtype=np.complex128)
+    expect[0] = expect[7] = S2
+    np.testing.assert_allclose(sv, expect, atol=1e-15)
+
+
+def test_hadamards_amplitude():
+    from tnc_amd import Circuit, TensorData
+
+    c = Circuit()
+    qr = c.allocate_register(5)
+    for q in qr.qubits():
+        c.append_gate(TensorData.from_gate("h"), [q])
+    tn, _ = c.into_amplitude_network("00000")
+    legs, data, ref = gpu_contract(tn)
+    assert legs == []
+    np.testing.assert_allclose(data, S2**5, atol=1e-15)
+
+
+def test_qft_expectation():
+    from tnc_amd import Circuit, RandomGreedy, TensorData
+
+    c = Circuit()
+    q = c.allocate_register(2)
+    c.append_gate(TensorData.from_gate("h"), [q.qubit(1)])
+    c.append_gate(TensorData.from_gate("cx"), [q.qubit(1), q.qubit(0)])
+    c.append_gate(TensorData.from_gate("h"), [q.qubit(1)])
+    c.append_gate(TensorData.from_gate("cp", [math.pi / 2]), [q.qubit(1), q.qubit(0)])
+    c.append_gate(TensorData.from_gate("h"), [q.qubit(0)])
+    c.append_gate(TensorData.from_gate("swap"), [q.qubit(0), q.qubit(1)])
+    tn = c.into_expectation_value_network()
+    legs, data, ref = gpu_contract(tn, RandomGreedy(3))
+    np.testing.assert_allclose(data, 0.5, atol=1e-15)
+
+
+def test_rqc24_fixture_vs_oracle():
+    """Config 2: the frozen 24q RQC amplitude network, frozen path."""
+    from tnc_amd.contraction_path import ContractionPath
+    from tnc_amd.executor import contract_tensor_network_gpu
+    from tnc_amd.fixtures import load_fixture
+
+    tn, replace_toplevel, meta = load_fixture("rqc24")
+    replace = ContractionPath.simple(replace_toplevel)
+    legs, data = contract_tensor_network_gpu(tn, replace)
+    ref = contract_network(network_to_otensors(tn), replace)
+    assert legs == ref.legs
+    # single amplitude: scalar; 1e-10 relative bar from the north star
+    np.testing.assert_allclose(data, ref.data, rtol=1e-10)
+
+
+def test_engine_repeatable():
+    """Leaves persist; repeated contract gives identical results."""
+    from tnc_amd import Greedy
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.executor import ContractionEngine
+
+    tn = random_circuit(12, 8, 0.5, 0.5, 5, ConnectivityLayout.EAGLE)
+    replace = Greedy().find_path(tn).replace_path()
+    eng = ContractionEngine(tn, replace)
+    try:
+        eng.contract()
+        _, first = eng.result()
+        eng.contract()
+        _, second = eng.result()
+        np.testing.assert_array_equal(first, second)
+        ref = contract_network(network_to_otensors(tn), replace)
+        np.testing.assert_allclose(first, ref.data, rtol=1e-10)
+    finally:
+        eng.close()
+
+
+def test_engine_profiled():
+    from tnc_amd import Greedy
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.executor import ContractionEngine
+
+    tn = random_circuit(10, 6, 0.5, 0.5, 9, ConnectivityLayout.EAGLE)
+    replace = Greedy().find_path(tn).replace_path()
+    eng = ContractionEngine(tn, replace)
+    try:
+        elapsed, step_ms, gemm_ms, kinds = eng.contract_profiled()
+        assert len(step_ms) == len(eng.steps)
+        assert all(m >= 0.0 for m in step_ms)
+        assert elapsed > 0
+    finally:
+        eng.close()
+
+
+def test_fails_loudly_without_extension(monkeypatch):
+    """The product path must never fall back: a missing .so raises."""
+    import importlib
+
+    import tnc_amd.hiplib as hiplib
+
+    monkeypatch.setattr(hiplib, "_LIB_PATH", "/nonexistent/libtnc_hip.so")
+    monkeypatch.setattr(hiplib, "_lib", None)
+    with pytest.raises(RuntimeError, match="not built"):
+        hiplib.lib()

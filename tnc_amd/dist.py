@@ -1,0 +1,144 @@
+"""Partition-parallel distributed contraction: one partition per GPU,
+RCCL send/recv of open-leg intermediates over xGMI.
+
+Mirrors tnc/src/mpi/communication.rs:
+- rank<->partition mapping: the partition holding the final tensor goes to
+  rank 0, the rest in order (get_tensor_mapping, communication.rs:89-115);
+- local phase: each rank contracts its partition independently (zero
+  communication);
+- fan-in (intermediate_reduce_tensor_network, communication.rs:199-249): for
+  each toplevel pair (x, y), rank(y) sends its contracted leaf to rank(x),
+  which contracts {local, received} as a single pair; the final tensor is
+  forwarded to rank 0 (communication.rs:236-247).
+
+Unlike the reference there is NO serialization: every rank derives the full
+plan (partitioning, paths, all intermediate leg lists and shapes)
+deterministically from the shared fixture, so the wire carries only raw c128
+device buffers. Tensors travel as float64 views (re, im interleaved), which
+both the gloo (CPU tests) and nccl/RCCL backends support.
+
+The contraction backends are injected so the orchestration is testable on
+CPU: the GPU backend lives in bench.py / executor.py; tests inject an
+oracle-based one.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Dict, List, Tuple
+
+from .contraction_path import ContractionPath
+from .cost import contract_cost_tensors
+from .partition import find_partitioning, partition_tensor_network
+from .paths import RandomGreedy
+from .tensor import CompositeTensor, LeafTensor
+
+
+class DistPlan:
+    """Deterministic distributed plan shared by all ranks."""
+
+    def __init__(self, partitioned: CompositeTensor, path: ContractionPath,
+                 nranks: int):
+        assert len(partitioned.tensors) <= nranks or True
+        self.partitioned = partitioned
+        self.path = path
+        self.nparts = len(partitioned.tensors)
+        # rank mapping (communication.rs:89-115): final partition -> rank 0
+        if path.toplevel:
+            final_part = path.toplevel[-1][0]
+        else:
+            final_part = 0
+        self.part_rank: Dict[int, int] = {final_part: 0}
+        nxt = 1
+        for idx in sorted(path.nested.keys()):
+            if idx != final_part:
+                self.part_rank[idx] = nxt
+                nxt += 1
+        for idx in range(self.nparts):  # parts without nested path (1 leaf)
+            if idx not in self.part_rank:
+                self.part_rank[idx] = nxt
+                nxt += 1
+        assert nxt <= nranks, f"need {nxt} ranks, got {nranks}"
+        self.used_ranks = nxt
+        # external (open-leg) view of each partition, and the evolution of
+        # leg lists through the fan-in — all statically known
+        self.externals: List[LeafTensor] = [
+            t.external_tensor() if isinstance(t, CompositeTensor) else t
+            for t in partitioned.tensors
+        ]
+        # fan-in flops (metric numerator share of the exchange phase)
+        views = [LeafTensor(t.legs, t.bond_dims) for t in self.externals]
+        self.fanin_flops = 0.0
+        self.fanin_shapes: List[Tuple[List[int], List[int]]] = []
+        for x, y in path.toplevel:
+            self.fanin_flops += contract_cost_tensors(views[x], views[y])
+            out = views[x] ^ views[y]
+            self.fanin_shapes.append((list(views[y].legs), list(views[y].bond_dims)))
+            views[x] = out
+        self.final_legs = views[path.toplevel[-1][0]].legs if path.toplevel else (
+            self.externals[0].legs)
+
+    def local_flops(self, part: int) -> float:
+        sub = self.partitioned.tensors[part]
+        if not isinstance(sub, CompositeTensor):
+            return 0.0
+        inner = self.path.nested.get(part)
+        if inner is None:
+            return 0.0
+        views = [LeafTensor(t.legs, t.bond_dims) for t in sub.tensors]
+        total = 0.0
+        for i, j in inner.toplevel:
+            total += contract_cost_tensors(views[i], views[j])
+            views[i] = views[i] ^ views[j]
+        return total
+
+    def total_flops(self) -> float:
+        return sum(self.local_flops(p) for p in range(self.nparts)) + self.fanin_flops
+
+
+def make_plan(tn: CompositeTensor, nranks: int, trials: int = 16,
+              size_cap=None, seed: int = 0) -> DistPlan:
+    """Partition + per-partition paths + fan-in path, all deterministic."""
+    if nranks == 1:
+        raise ValueError("use the single-GPU engine for one rank")
+    partitioning = find_partitioning(tn, nranks, seed=seed)
+    ptn = partition_tensor_network(tn, partitioning)
+    result = RandomGreedy(trials, size_cap=size_cap).find_path(ptn)
+    return DistPlan(ptn, result.replace_path(), nranks)
+
+
+def run_fanin(plan: DistPlan, rank: int,
+              local: "object",
+              send: Callable[[object, List[int], int], None],
+              recv: Callable[[List[int], List[int], int], object],
+              contract_pair: Callable[[object, List[int], List[int],
+                                       object, List[int], List[int]], object]):
+    """The fan-in walk (communication.rs:199-249). `local` is this rank's
+    contracted partition handle (backend-specific). Returns the final handle
+    on rank 0 (None elsewhere).
+
+    send(handle, legs, dims, peer); recv(legs, dims, peer) -> handle;
+    contract_pair(a, a_legs, a_dims, b, b_legs, b_dims) -> handle.
+    """
+    views = [LeafTensor(t.legs, t.bond_dims) for t in plan.externals]
+    final_rank = 0
+    for x, y in plan.path.toplevel:
+        receiver = plan.part_rank[x]
+        sender = plan.part_rank[y]
+        final_rank = receiver
+        if receiver == rank:
+            received = recv(views[y].legs, views[y].bond_dims, sender)
+            local = contract_pair(local, views[x].legs, views[x].bond_dims,
+                                  received, views[y].legs, views[y].bond_dims)
+        if sender == rank:
+            send(local, views[y].legs, views[y].bond_dims, receiver)
+            local = None
+        views[x] = views[x] ^ views[y]
+    # forward final tensor to rank 0 (communication.rs:236-247)
+    if final_rank != 0:
+        fx = plan.path.toplevel[-1][0] if plan.path.toplevel else 0
+        if rank == final_rank:
+            send(local, views[fx].legs, views[fx].bond_dims, 0)
+            local = None
+        elif rank == 0:
+            local = recv(views[fx].legs, views[fx].bond_dims, final_rank)
+    return local if rank == 0 else None

@@ -184,17 +184,23 @@ class RandomGreedy(_CotengrustLike):
     [0.001, 1]); keeps the path with the lowest op count. Seeded (default 42,
     like cotengrust.rs:71)."""
 
-    def __init__(self, ntrials, seed=42):
+    def __init__(self, ntrials, seed=42, size_cap=None):
         self.ntrials = ntrials
         self.seed = seed
+        # optional peak-memory cap (elements, contract_size_tensors metric):
+        # among trials within the cap, the lowest op count wins; with none
+        # within the cap, the smallest peak wins. The reference has no such
+        # cap (its CPU runs page instead); on a 288 GB GPU it is load-bearing.
+        self.size_cap = size_cap
 
     def _ctor_args(self):
-        return {"ntrials": self.ntrials, "seed": self.seed}
+        return {"ntrials": self.ntrials, "seed": self.seed,
+                "size_cap": self.size_cap}
 
     def _optimize_single(self, leaves, external):
         rng = np.random.Generator(np.random.PCG64(self.seed))
         best_path = None
-        best_cost = math.inf
+        best_key = None
         # trial 0 is the deterministic greedy; keeps RandomGreedy >= Greedy
         for trial in range(max(1, self.ntrials)):
             if trial == 0:
@@ -203,19 +209,28 @@ class RandomGreedy(_CotengrustLike):
                 costmod = math.exp(rng.uniform(math.log(0.1), math.log(4.0)))
                 temperature = math.exp(rng.uniform(math.log(0.001), math.log(1.0)))
             ssa = _greedy_ssa(leaves, costmod, temperature, rng)
-            cost = _ssa_op_cost(leaves, ssa)
-            if cost < best_cost:
-                best_cost = cost
+            cost, peak = _ssa_op_cost(leaves, ssa)
+            if self.size_cap is not None:
+                key = (peak > self.size_cap, peak if peak > self.size_cap else cost)
+            else:
+                key = (False, cost)
+            if best_key is None or key < best_key:
+                best_key = key
                 best_path = ssa
         return best_path or []
 
 
 def _ssa_op_cost(leaves, ssa_path):
-    """Op count (prod of union dims per step) of an SSA path over leaves."""
+    """(op count, peak size) of an SSA path over leaves — the op count is
+    prod-of-union-dims per step (contraction_cost.rs:49-52), the peak is
+    out+a+b elements (contraction_cost.rs:69-72)."""
     views = list(leaves)
     cost = 0.0
+    peak = 0.0
     for i, j in ssa_path:
         ti, tj = views[i], views[j]
         cost += (ti | tj).size()
-        views.append(ti ^ tj)
-    return cost
+        out = ti ^ tj
+        peak = max(peak, out.size() + ti.size() + tj.size())
+        views.append(out)
+    return cost, peak
