@@ -129,10 +129,29 @@ def test_strided_views():
 
     rng = np.random.default_rng(11)
     a = _rand((8, 12, 6), rng).transpose(2, 0, 1)  # non-contiguous view
-    b = _rand((12, 10), rng)[::1, :]
-    ref = oracle.contract_ndarrays([0, 1], [2, 0, 3], a, [3, 1], b)
-    got = hiplib.einsum_c128([0, 1], [2, 0, 3], a, [3, 1], b)
+    b = _rand((12, 10), rng)
+    # symdiff out order: A-only [2, 0] then B-only [1]
+    ref = oracle.contract_ndarrays([2, 0, 1], [2, 0, 3], a, [3, 1], b)
+    got = hiplib.einsum_c128([2, 0, 1], [2, 0, 3], a, [3, 1], b)
     np.testing.assert_allclose(got, ref, rtol=1e-12)
+    # strided through the GEMM path too
+    a2 = _rand((128, 96), rng).T  # labels [0, 1], shape (96, 128), strided
+    b2 = _rand((128, 80), rng)
+    ref = oracle.contract_ndarrays([0, 2], [0, 1], a2, [1, 2], b2)
+    got = hiplib.einsum_c128([0, 2], [0, 1], a2, [1, 2], b2)
+    np.testing.assert_allclose(got, ref, rtol=1e-11)
+
+
+def test_trace_rejected():
+    """Labels absent from both out and the other operand are outside the
+    tensor_mult boundary contract -> loud error, not silent garbage."""
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(12)
+    a = _rand((4, 3), rng)
+    b = _rand((3, 5), rng)
+    with pytest.raises(RuntimeError, match="neither contracted nor in out"):
+        hiplib.einsum_c128([2], [0, 1], a, [1, 2], b)  # leg 0 traced
 
 
 def test_skinny_anyk():

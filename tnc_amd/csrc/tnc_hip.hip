@@ -300,8 +300,9 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
     ci[f] = v4d{0.0, 0.0, 0.0, 0.0};
   }
 
-  // v_mfma_f64_16x16x4_f64 operand map: lane l holds A[i = l%16][k = l/16]
-  // and B[k = l/16][j = l%16]; D: lane l, reg r -> row (l/16)*4 + r, col l%16.
+  // v_mfma_f64_16x16x4_f64 operand map (verified on hardware,
+  // scripts/mfma_repro.hip): lane l holds A[i = l%16][k = l/16] and
+  // B[k = l/16][j = l%16]; D: lane l, reg r -> row 4*r + l/16, col l%16.
   const int fi = lane % 16;
   const int fk = lane / 16;
 
@@ -339,11 +340,11 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
     __syncthreads();
   }
 
-  const int crow0 = wave * 16 + (lane / 16) * 4;
+  const int crow0 = wave * 16 + (lane / 16);
   const int ccol = lane % 16;
   for (int f = 0; f < 4; ++f) {
     for (int r = 0; r < 4; ++r) {
-      u64 row = brow + crow0 + r;
+      u64 row = brow + crow0 + 4 * r;
       u64 col = bcol + f * 16 + ccol;
       if (interior || (row < M && col < N))
         C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
@@ -375,19 +376,28 @@ static int log2_u64(u64 v) {
   return s;
 }
 
-static int build_map(const std::vector<AxisInfo>& axes, GatherMap* m) {
+static bool axes_pow2(const std::vector<AxisInfo>& axes) {
+  for (const auto& a : axes)
+    if (a.dim & (a.dim - 1)) return false;
+  return true;
+}
+
+// Encode a map. The pow2 (shift/mask) and general (div/mod) ENCODINGS are
+// incompatible; when a kernel takes several maps the caller must pick ONE
+// encoding for all of them (allow_pow2 = the joint decision).
+static int build_map(const std::vector<AxisInfo>& axes, GatherMap* m,
+                     bool allow_pow2 = true) {
   int n = (int)axes.size();
   if (n > TN_MAXR) return -1;
   m->n = n;
   u64 pstride = 1;
-  bool p2 = true;
+  bool p2 = allow_pow2 && axes_pow2(axes);
   for (int i = n - 1; i >= 0; --i) {
     m->pstride[i] = pstride;
     m->dim[i] = axes[i].dim;
     m->sa[i] = axes[i].sa;
     m->sb[i] = axes[i].sb;
     pstride *= axes[i].dim;
-    if (axes[i].dim & (axes[i].dim - 1)) p2 = false;
   }
   m->pow2 = p2 ? 1 : 0;
   if (p2) {
@@ -481,6 +491,22 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       k_b.push_back(j);
     }
   }
+  // every input label must be contracted (shared) or appear in out — traces
+  // and repeated labels are outside the tensor_mult boundary contract
+  // (contraction.rs:88-116 always passes the symmetric difference)
+  auto in_out = [&](u64 lab) {
+    for (int o = 0; o < out_nd; ++o)
+      if (out_labels[o] == lab) return true;
+    return false;
+  };
+  for (int i = 0; i < A.nd; ++i)
+    if (find(B, A.labels[i]) < 0 && !in_out(A.labels[i]))
+      FAILV(TN_ERR_INVALID, "label %llu of A neither contracted nor in out",
+            (unsigned long long)A.labels[i]);
+  for (int i = 0; i < B.nd; ++i)
+    if (find(A, B.labels[i]) < 0 && !in_out(B.labels[i]))
+      FAILV(TN_ERR_INVALID, "label %llu of B neither contracted nor in out",
+            (unsigned long long)B.labels[i]);
   u64 nout = M * N;
   if (stats) {
     stats->m = M;
@@ -529,10 +555,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     for (size_t t = 0; t < k_a.size(); ++t)
       kax.push_back({A.dims[k_a[t]], A.strides[k_a[t]], B.strides[k_b[t]]});
     if (kax.empty()) kax.push_back({1, 0, 0});
+    bool p2 = axes_pow2(oax) && axes_pow2(kax);
     GatherMap omap, kmap;
-    if (build_map(oax, &omap) || build_map(kax, &kmap))
+    if (build_map(oax, &omap, p2) || build_map(kax, &kmap, p2))
       FAILV(TN_ERR_INVALID, "rank too large");
-    bool p2 = omap.pow2 && kmap.pow2;
     int blocks = grid_for(nout);
     if (K <= TN_SMALLK) {
       if (p2)
