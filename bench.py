@@ -90,7 +90,7 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds):
     }
 
 
-def cpu_baseline(infos, budget_s=15.0, cap_elems=2 ** 27):
+def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29):
     """Oracle (numpy einsum -> BLAS zgemm) timed on the host cores over a
     bounded sample: the largest path steps whose operands fit `cap_elems`,
     random-valued inputs of the same shapes (einsum time is value-
@@ -99,6 +99,7 @@ def cpu_baseline(infos, budget_s=15.0, cap_elems=2 ** 27):
 
     import oracle
 
+    all_flops = sum(i.flops for i in infos)
     order = sorted(range(len(infos)), key=lambda s: -infos[s].flops)
     rng = np.random.default_rng(0)
     total_flops = 0.0
@@ -134,8 +135,10 @@ def cpu_baseline(infos, budget_s=15.0, cap_elems=2 ** 27):
         "unit": "GFLOP/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"{used} largest rqc36 path steps with operands <= "
-                  f"{cap_elems} elems, random-valued, {total_time:.1f}s",
+        "sample": f"{used} largest {fixture} path steps with operands <= "
+                  f"{cap_elems} elems ({100*total_flops/all_flops:.1f}% of "
+                  f"path flops), random-valued same-shape zgemm, "
+                  f"{total_time:.1f}s",
     }
 
 
@@ -165,7 +168,7 @@ def run_single(args):
     wall = time.perf_counter() - t0
 
     value = flops_per_contraction * args.steps / wall / 1e9
-    cb = None if args.no_cpu_baseline else cpu_baseline(eng.infos)
+    cb = None if args.no_cpu_baseline else cpu_baseline(eng.infos, args.fixture)
     eng.close()
     emit({
         "metric": "pairwise-contraction GFLOP/s (c128)",

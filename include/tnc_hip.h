@@ -78,6 +78,12 @@ typedef struct tn_net tn_net;
 /* Create an executor bound to `device`. */
 tn_net* tn_net_create(int device);
 
+/* Reserve a device arena of `bytes` for intermediates and packing
+ * workspaces (one hipMalloc; first-fit, stream-ordered reuse). Without a
+ * reservation the executor falls back to stream-ordered hipMallocAsync,
+ * which costs seconds per contraction at tens-of-GB intermediates. */
+int tn_net_reserve(tn_net* net, uint64_t bytes);
+
 /* Register leaf `index = return value` with labels/dims and host data
  * (complex128, row-major, contiguous). Data is uploaded immediately and
  * persists across contract calls. Returns a negative value on error. */

@@ -165,3 +165,13 @@ def test_high_rank():
     a_labels = list(range(16))
     b_labels = list(range(8, 24))
     run_case(a_labels, [2] * 16, b_labels, [2] * 16, rtol=1e-11)
+
+
+def test_gemm_splitk():
+    # few tiles + deep K -> split-K partials + reduce
+    run_case([0, 1], [64, 4096], [1, 2], [4096, 64], rtol=1e-11)
+    run_case([0, 1], [128, 65536], [1, 2], [65536, 128], rtol=1e-10)
+    # deep-K with multiple qubit legs (pow2 maps + packing)
+    a_labels = [0, 1] + list(range(10, 22))
+    b_labels = list(range(10, 22)) + [2, 3]
+    run_case(a_labels, [2] * 14, b_labels, [2] * 14, rtol=1e-10)

@@ -225,25 +225,30 @@ __global__ void k_permute_c128(const double2* __restrict__ src,
 
 __global__ __launch_bounds__(256) void k_zgemm_v1(
     const double2* __restrict__ A, const double2* __restrict__ B,
-    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles) {
+    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
+    u64 kchunk) {
   __shared__ double2 As[GT][GK + 1];
   __shared__ double2 Bs[GK][GT + 1];
   const int tx = threadIdx.x % 16, ty = threadIdx.x / 16;
-  const u64 tile = blockIdx.x;
+  const u64 tile = blockIdx.x % tiles;
+  const u64 slice = blockIdx.x / tiles;
   const u64 brow = (tile / col_tiles) * GT, bcol = (tile % col_tiles) * GT;
+  const u64 kbeg = slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += slice * M * N;  // slice 0 == C itself when kchunk == K
   double2 acc[4][4];
   for (int i = 0; i < 4; ++i)
     for (int j = 0; j < 4; ++j) acc[i][j] = make_double2(0.0, 0.0);
-  for (u64 k0 = 0; k0 < K; k0 += GK) {
+  for (u64 k0 = kbeg; k0 < kend; k0 += GK) {
     for (int i = threadIdx.x; i < GT * GK; i += 256) {
       int r = i / GK, c = i % GK;
-      As[r][c] = (brow + r < M && k0 + c < K) ? A[(brow + r) * K + k0 + c]
-                                              : make_double2(0.0, 0.0);
+      As[r][c] = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
+                                                 : make_double2(0.0, 0.0);
     }
     for (int i = threadIdx.x; i < GK * GT; i += 256) {
       int r = i / GT, c = i % GT;
-      Bs[r][c] = (k0 + r < K && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
-                                              : make_double2(0.0, 0.0);
+      Bs[r][c] = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
+                                                 : make_double2(0.0, 0.0);
     }
     __syncthreads();
     for (int kk = 0; kk < GK; ++kk) {
@@ -282,7 +287,8 @@ typedef double v4d __attribute__((ext_vector_type(4)));
 
 __global__ __launch_bounds__(256) void k_zgemm_mfma(
     const double2* __restrict__ A, const double2* __restrict__ B,
-    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles) {
+    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
+    u64 kchunk) {
   __shared__ double Ar[MF_T * A_LD];
   __shared__ double Ai[MF_T * A_LD];
   __shared__ double Br[MF_K * B_LD];
@@ -290,8 +296,12 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
-  const u64 tile = blockIdx.x;
+  const u64 tile = blockIdx.x % tiles;
+  const u64 slice = blockIdx.x / tiles;
   const u64 brow = (tile / col_tiles) * MF_T, bcol = (tile % col_tiles) * MF_T;
+  const u64 kbeg = slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += slice * M * N;
   const bool interior = (brow + MF_T <= M) && (bcol + MF_T <= N);
 
   v4d cr[4], ci[4];
@@ -306,18 +316,18 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
   const int fi = lane % 16;
   const int fk = lane / 16;
 
-  for (u64 k0 = 0; k0 < K; k0 += MF_K) {
+  for (u64 k0 = kbeg; k0 < kend; k0 += MF_K) {
     for (int i = threadIdx.x; i < MF_T * MF_K; i += 256) {
       int r = i / MF_K, c = i % MF_K;
-      double2 v = (brow + r < M && k0 + c < K) ? A[(brow + r) * K + k0 + c]
-                                               : make_double2(0.0, 0.0);
+      double2 v = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
+                                                  : make_double2(0.0, 0.0);
       Ar[r * A_LD + c] = v.x;
       Ai[r * A_LD + c] = v.y;
     }
     for (int i = threadIdx.x; i < MF_K * MF_T; i += 256) {
       int r = i / MF_T, c = i % MF_T;
-      double2 v = (k0 + r < K && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
-                                               : make_double2(0.0, 0.0);
+      double2 v = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
+                                                  : make_double2(0.0, 0.0);
       Br[r * B_LD + c] = v.x;
       Bi[r * B_LD + c] = v.y;
     }
@@ -349,6 +359,22 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
       if (interior || (row < M && col < N))
         C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
     }
+  }
+}
+
+// split-K reduce: C[p] = sum over slices of ws[s][p]
+__global__ void k_splitk_reduce(const double2* __restrict__ ws,
+                                double2* __restrict__ C, u64 nout,
+                                int slices) {
+  for (u64 p = blockIdx.x * (u64)blockDim.x + threadIdx.x; p < nout;
+       p += gridDim.x * (u64)blockDim.x) {
+    double re = 0.0, im = 0.0;
+    for (int s = 0; s < slices; ++s) {
+      double2 v = ws[(u64)s * nout + p];
+      re += v.x;
+      im += v.y;
+    }
+    C[p] = make_double2(re, im);
   }
 }
 
@@ -429,6 +455,99 @@ struct StepStats {
   hipEvent_t gemm_ev1 = nullptr;
 };
 
+// --- device arena: one reserved slab, host-side first-fit bookkeeping ---
+// Reuse is stream-ordered (all users launch on one stream), so a freed
+// block can be handed out immediately: the consuming kernels are ordered
+// behind the producing ones. Avoids hipMallocAsync's per-allocation page
+// mapping, which costs seconds per contraction at 30-70 GB intermediates.
+struct Arena {
+  char* base = nullptr;
+  size_t size = 0;
+  struct Block {
+    size_t off, sz;
+    bool free;
+  };
+  std::vector<Block> blocks;
+
+  int reserve(size_t bytes) {
+    if (base) return 0;
+    if (hipMalloc((void**)&base, bytes) != hipSuccess) return -1;
+    size = bytes;
+    blocks = {{0, bytes, true}};
+    return 0;
+  }
+  void* alloc(size_t bytes) {
+    if (!base) return nullptr;
+    bytes = (bytes + 255) & ~(size_t)255;
+    for (size_t i = 0; i < blocks.size(); ++i) {
+      if (blocks[i].free && blocks[i].sz >= bytes) {
+        size_t off = blocks[i].off;
+        if (blocks[i].sz > bytes) {
+          blocks.insert(blocks.begin() + i + 1,
+                        {off + bytes, blocks[i].sz - bytes, true});
+        }
+        blocks[i].sz = bytes;
+        blocks[i].free = false;
+        return base + off;
+      }
+    }
+    return nullptr;  // exhausted -> caller falls back
+  }
+  bool owns(const void* p) const {
+    return base && p >= base && p < base + size;
+  }
+  void release(void* p) {
+    size_t off = (char*)p - base;
+    for (size_t i = 0; i < blocks.size(); ++i) {
+      if (blocks[i].off == off) {
+        blocks[i].free = true;
+        // merge with next, then previous
+        if (i + 1 < blocks.size() && blocks[i + 1].free) {
+          blocks[i].sz += blocks[i + 1].sz;
+          blocks.erase(blocks.begin() + i + 1);
+        }
+        if (i > 0 && blocks[i - 1].free) {
+          blocks[i - 1].sz += blocks[i].sz;
+          blocks.erase(blocks.begin() + i);
+        }
+        return;
+      }
+    }
+  }
+  void destroy() {
+    if (base) (void)hipFree(base);
+    base = nullptr;
+    blocks.clear();
+  }
+};
+
+// workspace allocation context threaded through the einsum implementation
+struct WsCtx {
+  Arena* arena = nullptr;  // optional
+  hipStream_t stream = nullptr;
+};
+
+static int ws_alloc(WsCtx& ctx, void** p, size_t bytes) {
+  if (ctx.arena) {
+    *p = ctx.arena->alloc(bytes);
+    if (*p) return TN_OK;
+  }
+  if (hipMallocAsync(p, bytes, ctx.stream) != hipSuccess) {
+    g_last_error = "device allocation failed";
+    return TN_ERR_OOM;
+  }
+  return TN_OK;
+}
+
+static void ws_free(WsCtx& ctx, void* p) {
+  if (!p) return;
+  if (ctx.arena && ctx.arena->owns(p)) {
+    ctx.arena->release(p);
+    return;
+  }
+  (void)hipFreeAsync(p, ctx.stream);
+}
+
 static int grid_for(u64 nout, int block = 256) {
   u64 blocks = (nout + (u64)block - 1) / block;
   u64 cap = 64 * 2048;  // grid-stride covers the remainder
@@ -440,7 +559,7 @@ static int grid_for(u64 nout, int block = 256) {
 // core einsum over device buffers; out is contiguous row-major in out order.
 static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
                            int out_nd, const Meta& A, const Meta& B,
-                           double2* out, hipStream_t stream,
+                           double2* out, hipStream_t stream, WsCtx& ws,
                            StepStats* stats) {
   if (A.nd > TN_MAXR || B.nd > TN_MAXR || out_nd > TN_MAXR)
     FAILV(TN_ERR_INVALID, "tensor rank exceeds %d", TN_MAXR);
@@ -525,16 +644,19 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (build_map(kax, &kmap)) FAILV(TN_ERR_INVALID, "rank too large");
     int blocks = grid_for(K);
     if (blocks > 2048) blocks = 2048;
-    double2* ws;
-    HIP_CHECK(hipMallocAsync((void**)&ws, blocks * sizeof(double2), stream));
+    double2* wsbuf;
+    {
+      int rc_ = ws_alloc(ws, (void**)&wsbuf, blocks * sizeof(double2));
+      if (rc_) return rc_;
+    }
     if (kmap.pow2)
-      k_dot_partial<true><<<blocks, 256, 0, stream>>>(A.data, B.data, ws, K,
-                                                      kmap);
+      k_dot_partial<true><<<blocks, 256, 0, stream>>>(A.data, B.data, wsbuf,
+                                                      K, kmap);
     else
-      k_dot_partial<false><<<blocks, 256, 0, stream>>>(A.data, B.data, ws, K,
-                                                       kmap);
-    k_dot_finish<<<1, 256, 0, stream>>>(ws, out, blocks);
-    HIP_CHECK(hipFreeAsync(ws, stream));
+      k_dot_partial<false><<<blocks, 256, 0, stream>>>(A.data, B.data, wsbuf,
+                                                       K, kmap);
+    k_dot_finish<<<1, 256, 0, stream>>>(wsbuf, out, blocks);
+    ws_free(ws, wsbuf);
     HIP_CHECK(hipGetLastError());
     return TN_OK;
   }
@@ -612,7 +734,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     GatherMap map;
     if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = M * K;
-    HIP_CHECK(hipMallocAsync((void**)&packA, elems * sizeof(double2), stream));
+    {
+      int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(double2));
+      if (rc_) return rc_;
+    }
     int blocks = grid_for(elems);
     if (map.pow2)
       k_permute_c128<true><<<blocks, 256, 0, stream>>>(A.data, packA, elems,
@@ -628,7 +753,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     GatherMap map;
     if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = K * N;
-    HIP_CHECK(hipMallocAsync((void**)&packB, elems * sizeof(double2), stream));
+    {
+      int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(double2));
+      if (rc_) return rc_;
+    }
     int blocks = grid_for(elems);
     if (map.pow2)
       k_permute_c128<true><<<blocks, 256, 0, stream>>>(B.data, packB, elems,
@@ -647,23 +775,51 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   double2* tmpC = nullptr;
   if (!direct) {
     if (stats) stats->kind = 3;
-    HIP_CHECK(hipMallocAsync((void**)&tmpC, nout * sizeof(double2), stream));
+    {
+      int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(double2));
+      if (rc_) return rc_;
+    }
     Cg = tmpC;
   }
 
   u64 row_tiles = (M + GT - 1) / GT;
   u64 col_tiles = (N + GT - 1) / GT;
-  dim3 grid((unsigned)(row_tiles * col_tiles));
+  u64 tiles = row_tiles * col_tiles;
+  // split-K: with few tiles and deep K, slice K across extra blocks into
+  // partial buffers + a reduce pass (fills the 256 CUs; a 64-tile K=2^20
+  // GEMM goes from ~6 to ~50 TFLOP/s)
+  u64 splitk = 1;
+  if (tiles < 192 && K >= 256) {
+    while (tiles * splitk < 256 && K / (splitk * 2) >= 64) splitk *= 2;
+    if (splitk > 64) splitk = 64;
+  }
+  u64 kchunk = (K + splitk - 1) / splitk;
+  kchunk = ((kchunk + MF_K - 1) / MF_K) * MF_K;  // tile-aligned
+  splitk = (K + kchunk - 1) / kchunk;
+  double2* gemm_out = Cg;
+  double2* splitbuf = nullptr;
+  if (splitk > 1) {
+    int rc_ = ws_alloc(ws, (void**)&splitbuf, splitk * nout * sizeof(double2));
+    if (rc_) return rc_;
+    gemm_out = splitbuf;
+  }
+  dim3 grid((unsigned)(tiles * splitk));
   bool mfma = (M >= 32 && N >= 32);
   if (stats && stats->gemm_ev0)
     HIP_CHECK(hipEventRecord(stats->gemm_ev0, stream));
   if (mfma)
-    k_zgemm_mfma<<<grid, 256, 0, stream>>>(Ag, Bg, Cg, M, N, K, col_tiles);
+    k_zgemm_mfma<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
+                                           col_tiles, tiles, kchunk);
   else
-    k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, Cg, M, N, K, col_tiles);
+    k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K, col_tiles,
+                                         tiles, kchunk);
+  if (splitk > 1)
+    k_splitk_reduce<<<grid_for(nout), 256, 0, stream>>>(splitbuf, Cg, nout,
+                                                        (int)splitk);
   if (stats && stats->gemm_ev1)
     HIP_CHECK(hipEventRecord(stats->gemm_ev1, stream));
   HIP_CHECK(hipGetLastError());
+  if (splitbuf) ws_free(ws, splitbuf);
 
   if (!direct) {
     // permute tmp [M legs (out order)][N legs (out order)] -> out order
@@ -690,9 +846,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       k_permute_c128<false><<<blocks, 256, 0, stream>>>(tmpC, out, nout, map);
     HIP_CHECK(hipGetLastError());
   }
-  if (packA) HIP_CHECK(hipFreeAsync(packA, stream));
-  if (packB) HIP_CHECK(hipFreeAsync(packB, stream));
-  if (tmpC) HIP_CHECK(hipFreeAsync(tmpC, stream));
+  ws_free(ws, packA);
+  ws_free(ws, packB);
+  ws_free(ws, tmpC);
   return TN_OK;
 }
 
@@ -751,8 +907,9 @@ extern "C" int tn_einsum_c128_dev(const u64* out_labels, const u64* out_shape,
   if (rc) return rc;
   rc = fill_meta(&B, b_labels, b_shape, b_strides, b_dev, b_ndim);
   if (rc) return rc;
+  WsCtx ws{nullptr, (hipStream_t)stream};
   return einsum_dev_impl(out_labels, out_shape, (int)out_ndim, A, B,
-                         (double2*)out_dev, (hipStream_t)stream, nullptr);
+                         (double2*)out_dev, (hipStream_t)stream, ws, nullptr);
 }
 
 static u64 span_elems(const u64* shape, const i64* strides, size_t nd) {
@@ -824,7 +981,9 @@ struct tn_net {
   hipStream_t stream = nullptr;
   std::vector<DevTensor> leaves;
   DevTensor final_t;       // final tensor of the last contract (owned)
+  bool final_in_arena = false;
   bool has_final = false;
+  Arena arena;
   u64 pool_in_use = 0;
 };
 
@@ -846,6 +1005,15 @@ extern "C" tn_net* tn_net_create(int device) {
     return nullptr;
   }
   return net;
+}
+
+extern "C" int tn_net_reserve(tn_net* net, uint64_t bytes) {
+  if (!net) FAILV(TN_ERR_INVALID, "null net");
+  HIP_CHECK(hipSetDevice(net->device));
+  if (net->arena.reserve(bytes))
+    FAILV(TN_ERR_OOM, "arena reservation of %llu bytes failed",
+          (unsigned long long)bytes);
+  return TN_OK;
 }
 
 extern "C" int64_t tn_net_add_leaf(tn_net* net, const u64* labels,
@@ -924,10 +1092,15 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
                          double* elapsed_ms) {
   if (!net) FAILV(TN_ERR_INVALID, "null net");
   HIP_CHECK(hipSetDevice(net->device));
+  WsCtx ws{net->arena.base ? &net->arena : nullptr, net->stream};
   if (net->has_final && net->final_t.owned) {
-    HIP_CHECK(hipFree(net->final_t.data));
+    if (net->final_in_arena)
+      net->arena.release(net->final_t.data);
+    else
+      (void)hipFree(net->final_t.data);
     net->final_t = DevTensor();
     net->has_final = false;
+    net->final_in_arena = false;
   }
   size_t n = net->leaves.size();
   std::vector<DevTensor> slots(net->leaves);  // shallow copies; owned=false
@@ -964,10 +1137,9 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     symdiff(A, B, &out.labels, &out.dims);
     out.elems = 1;
     for (u64 d : out.dims) out.elems *= d;
-    if (hipMallocAsync((void**)&out.data, out.elems * 16, net->stream) !=
-        hipSuccess) {
+    if (ws_alloc(ws, (void**)&out.data, out.elems * 16) != TN_OK) {
       rc = TN_ERR_OOM;
-      g_last_error = "hipMallocAsync failed for intermediate";
+      g_last_error = "device allocation failed for intermediate";
       break;
     }
     out.owned = true;
@@ -1003,17 +1175,17 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     }
     rc = einsum_dev_impl(out.labels.data(), out.dims.data(),
                          (int)out.labels.size(), ma, mb, out.data, net->stream,
-                         &st);
+                         ws, &st);
     if (profiled) HIP_CHECK(hipEventRecord(ev[4 * s + 1], net->stream));
     if (kind) kind[s] = st.kind;
     if (profiled) kinds_local[s] = st.kind;
     if (rc != TN_OK) {
-      (void)hipFreeAsync(out.data, net->stream);
+      ws_free(ws, out.data);
       break;
     }
     // free consumed intermediates (leaves persist)
-    if (A.owned) HIP_CHECK(hipFreeAsync(A.data, net->stream));
-    if (B.owned) HIP_CHECK(hipFreeAsync(B.data, net->stream));
+    if (A.owned) ws_free(ws, A.data);
+    if (B.owned) ws_free(ws, B.data);
     slots[i] = std::move(out);
     alive[j] = 0;
   }
@@ -1067,6 +1239,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
           net->has_final = true;
         }
       } else {
+        net->final_in_arena = ws.arena && ws.arena->owns(f.data);
         net->final_t = std::move(f);
         net->has_final = true;
       }
@@ -1076,7 +1249,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     // free any owned intermediates left over
     for (size_t x = 0; x < n; ++x)
       if (alive[x] && slots[x].owned && slots[x].data)
-        (void)hipFreeAsync(slots[x].data, net->stream);
+        ws_free(ws, slots[x].data);
     (void)hipStreamSynchronize(net->stream);
   }
   (void)hipEventDestroy(walk_start);
@@ -1148,7 +1321,9 @@ extern "C" void tn_net_destroy(tn_net* net) {
   (void)hipStreamSynchronize(net->stream);
   for (auto& t : net->leaves)
     if (t.data && !t.owned && !t.external) (void)hipFree(t.data);
-  if (net->has_final && net->final_t.owned) (void)hipFree(net->final_t.data);
+  if (net->has_final && net->final_t.owned && !net->final_in_arena)
+    (void)hipFree(net->final_t.data);
+  net->arena.destroy();
   (void)hipStreamDestroy(net->stream);
   delete net;
 }

@@ -48,6 +48,23 @@ def plan_steps(leaves, steps):
     return infos
 
 
+def arena_bytes(leaves, steps, infos):
+    """Peak device-arena demand: live intermediates + this step's output and
+    worst-case packing workspaces (A' + B' + unpacked C), walked over the
+    plan. 16 B per c128 element; padded 15% for first-fit fragmentation."""
+    nleaves = len(leaves)
+    live = {}  # slot -> bytes (intermediates only; leaves live outside)
+    peak = 0
+    for info in infos:
+        out_b = info.m * info.n * 16
+        extra = (info.m * info.k + info.k * info.n) * 16 + out_b  # packs+tmpC
+        demand = sum(live.values()) + out_b + extra
+        peak = max(peak, demand)
+        live.pop(info.j, None)
+        live[info.i] = out_b
+    return int(peak * 1.15) + (1 << 20)
+
+
 class ContractionEngine:
     """Device-resident executor for one (flattened) network."""
 
@@ -65,6 +82,18 @@ class ContractionEngine:
         self.net = L.tn_net_create(device)
         if not self.net:
             raise RuntimeError(f"tn_net_create failed: {hiplib.last_error()}")
+        reserve = arena_bytes(leaves, steps, self.infos)
+        # arena is an optimization: if the reservation fails (tiny GPUs,
+        # fragmented memory), the executor falls back to hipMallocAsync
+        if reserve > 64 * 1024 * 1024:
+            rc = L.tn_net_reserve(self.net, reserve)
+            if rc != 0:
+                import warnings
+
+                warnings.warn(
+                    f"arena reservation of {reserve} bytes failed "
+                    f"({hiplib.last_error()}); falling back to async allocs"
+                )
         for t in leaves:
             data = np.ascontiguousarray(t.tensordata.into_data(), dtype=np.complex128)
             assert list(data.shape) == list(t.bond_dims), (data.shape, t.bond_dims)

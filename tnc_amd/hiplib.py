@@ -47,6 +47,8 @@ def lib() -> ctypes.CDLL:
         ]
         L.tn_net_create.restype = ctypes.c_void_p
         L.tn_net_create.argtypes = [ctypes.c_int]
+        L.tn_net_reserve.restype = ctypes.c_int
+        L.tn_net_reserve.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
         L.tn_net_add_leaf.restype = ctypes.c_int64
         L.tn_net_add_leaf.argtypes = [
             ctypes.c_void_p, u64p, u64p, ctypes.c_size_t, ctypes.c_void_p,
