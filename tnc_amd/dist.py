@@ -33,6 +33,28 @@ from .paths import RandomGreedy
 from .tensor import CompositeTensor, LeafTensor
 
 
+def _local_final_view(part, inner: "ContractionPath | None") -> LeafTensor:
+    """Leg list/order of a partition's local contraction result: simulate
+    the replace-left walk over metadata (contraction.rs:52-57 semantics)."""
+    if not isinstance(part, CompositeTensor):
+        return LeafTensor(part.legs, part.bond_dims)
+    views = [
+        _local_final_view(t, inner.nested.get(i) if inner else None)
+        if isinstance(t, CompositeTensor)
+        else LeafTensor(t.legs, t.bond_dims)
+        for i, t in enumerate(part.tensors)
+    ]
+    if inner is None or not inner.toplevel:
+        assert len(views) == 1
+        return views[0]
+    for i, j in inner.toplevel:
+        views[i] = views[i] ^ views[j]
+        views[j] = None
+    remaining = [v for v in views if v is not None]
+    assert len(remaining) == 1
+    return remaining[0]
+
+
 class DistPlan:
     """Deterministic distributed plan shared by all ranks."""
 
@@ -59,11 +81,14 @@ class DistPlan:
                 nxt += 1
         assert nxt <= nranks, f"need {nxt} ranks, got {nranks}"
         self.used_ranks = nxt
-        # external (open-leg) view of each partition, and the evolution of
-        # leg lists through the fan-in — all statically known
+        # Open-leg view of each partition IN THE ORDER the local replace-left
+        # walk actually produces (simulated statically from the nested path)
+        # — the wire carries raw buffers, so leg order must be derived, not
+        # assumed (the reference ships leg metadata on the wire instead,
+        # serialization.rs; here the plan is deterministic on every rank).
         self.externals: List[LeafTensor] = [
-            t.external_tensor() if isinstance(t, CompositeTensor) else t
-            for t in partitioned.tensors
+            _local_final_view(t, path.nested.get(idx))
+            for idx, t in enumerate(partitioned.tensors)
         ]
         # fan-in flops (metric numerator share of the exchange phase)
         views = [LeafTensor(t.legs, t.bond_dims) for t in self.externals]
