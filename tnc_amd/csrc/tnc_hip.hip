@@ -273,19 +273,24 @@ __global__ __launch_bounds__(256) void k_zgemm_v1(
   }
 }
 
-// MFMA f64 kernel: v_mfma_f64_16x16x4_f64, 4 waves per block, 64x64 tile.
+// MFMA f64 kernel: v_mfma_f64_16x16x4_f64, 8 waves per block, 128x64 tile
+// (tile scan on the dominant rqc36 shape: 128x64/8-wave 58.7 TF/s vs
+// 64x64/4-wave 56.2; deeper K-tiles and 128-wide tiles lose to occupancy;
+// Gauss 3-mult loses to AGPR pressure — scripts/zgemm_tune.hip).
 // Wave w owns C rows [w*16, w*16+16); its row slab is 4 column fragments of
 // 16x16, each a {re, im} accumulator pair (4 f64 regs each). Per k-quad:
 // 4 MFMAs per fragment (Cr += ArBr; Cr += (-Ai)Bi; Ci += ArBi; Ci += AiBr).
 // LDS staging is planar with padded rows to avoid bank conflicts.
 typedef double v4d __attribute__((ext_vector_type(4)));
 
-#define MF_T 64
+#define MF_T 128      // tile rows = MF_WAVES * 16
+#define MF_TN 64      // tile cols
 #define MF_K 16
-#define A_LD 17  // padded row stride (doubles) for the 64x16 A tiles
-#define B_LD 66  // padded row stride for the 16x64 B tiles
+#define MF_THREADS 512
+#define A_LD 17  // padded row stride (doubles) for the A tiles
+#define B_LD 66  // padded row stride for the B tiles
 
-__global__ __launch_bounds__(256) void k_zgemm_mfma(
+__global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
     const double2* __restrict__ A, const double2* __restrict__ B,
     double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
     u64 kchunk) {
@@ -298,11 +303,11 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
   const int lane = threadIdx.x % 64;
   const u64 tile = blockIdx.x % tiles;
   const u64 slice = blockIdx.x / tiles;
-  const u64 brow = (tile / col_tiles) * MF_T, bcol = (tile % col_tiles) * MF_T;
+  const u64 brow = (tile / col_tiles) * MF_T, bcol = (tile % col_tiles) * MF_TN;
   const u64 kbeg = slice * kchunk;
   const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
   C += slice * M * N;
-  const bool interior = (brow + MF_T <= M) && (bcol + MF_T <= N);
+  const bool interior = (brow + MF_T <= M) && (bcol + MF_TN <= N);
 
   v4d cr[4], ci[4];
   for (int f = 0; f < 4; ++f) {
@@ -317,15 +322,15 @@ __global__ __launch_bounds__(256) void k_zgemm_mfma(
   const int fk = lane / 16;
 
   for (u64 k0 = kbeg; k0 < kend; k0 += MF_K) {
-    for (int i = threadIdx.x; i < MF_T * MF_K; i += 256) {
+    for (int i = threadIdx.x; i < MF_T * MF_K; i += MF_THREADS) {
       int r = i / MF_K, c = i % MF_K;
       double2 v = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
                                                   : make_double2(0.0, 0.0);
       Ar[r * A_LD + c] = v.x;
       Ai[r * A_LD + c] = v.y;
     }
-    for (int i = threadIdx.x; i < MF_K * MF_T; i += 256) {
-      int r = i / MF_T, c = i % MF_T;
+    for (int i = threadIdx.x; i < MF_K * MF_TN; i += MF_THREADS) {
+      int r = i / MF_TN, c = i % MF_TN;
       double2 v = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
                                                   : make_double2(0.0, 0.0);
       Br[r * B_LD + c] = v.x;
@@ -782,7 +787,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     Cg = tmpC;
   }
 
-  u64 row_tiles = (M + GT - 1) / GT;
+  bool mfma = (M >= 32 && N >= 32);
+  u64 row_tile_h = mfma ? MF_T : GT;
+  u64 row_tiles = (M + row_tile_h - 1) / row_tile_h;
   u64 col_tiles = (N + GT - 1) / GT;
   u64 tiles = row_tiles * col_tiles;
   // split-K: with few tiles and deep K, slice K across extra blocks into
@@ -804,12 +811,11 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     gemm_out = splitbuf;
   }
   dim3 grid((unsigned)(tiles * splitk));
-  bool mfma = (M >= 32 && N >= 32);
   if (stats && stats->gemm_ev0)
     HIP_CHECK(hipEventRecord(stats->gemm_ev0, stream));
   if (mfma)
-    k_zgemm_mfma<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
-                                           col_tiles, tiles, kchunk);
+    k_zgemm_mfma<<<grid, MF_THREADS, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
+                                                  col_tiles, tiles, kchunk);
   else
     k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K, col_tiles,
                                          tiles, kchunk);
