@@ -12,7 +12,7 @@ typedef unsigned long long u64;
 
 // Templated kernel: TM rows (= WAVES*16), 64 cols, KT-deep K tile.
 // REORDER: 0 = per-fragment 4-MFMA chain, 1 = pass-per-term (dep distance 4).
-template <int WAVES, int KT, int REORDER, int FRAGS = 4>
+template <int WAVES, int KT, int REORDER, int FRAGS = 4, int GROUPC = 0>
 __global__ __launch_bounds__(WAVES * 64) void zg(
     const double2* __restrict__ A, const double2* __restrict__ B,
     double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles) {
@@ -27,7 +27,20 @@ __global__ __launch_bounds__(WAVES * 64) void zg(
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
   const u64 tile = blockIdx.x;
-  const u64 brow = (tile / col_tiles) * TM, bcol = (tile % col_tiles) * TN;
+  u64 trow, tcol;
+  if (GROUPC > 0) {
+    const u64 row_tiles = gridDim.x / col_tiles;
+    const u64 g = tile / (row_tiles * GROUPC);
+    const u64 rem = tile - g * row_tiles * GROUPC;
+    const u64 wc = (GROUPC < col_tiles - g * GROUPC) ? GROUPC
+                                                     : col_tiles - g * GROUPC;
+    trow = rem / wc;
+    tcol = g * GROUPC + rem % wc;
+  } else {
+    trow = tile / col_tiles;
+    tcol = tile % col_tiles;
+  }
+  const u64 brow = trow * TM, bcol = tcol * TN;
   v4d cr[FRAGS], ci[FRAGS];
   v4d p3[REORDER == 2 ? FRAGS : 1];
   for (int f = 0; f < FRAGS; ++f) {
@@ -114,14 +127,15 @@ struct Variant {
   void (*launch)(const double2*, const double2*, double2*, u64, u64, u64);
 };
 
-template <int WAVES, int KT, int REORDER, int FRAGS = 4>
+template <int WAVES, int KT, int REORDER, int FRAGS = 4, int GROUPC = 0>
 static void launch_zg(const double2* A, const double2* B, double2* C, u64 M,
                       u64 N, u64 K) {
   constexpr int TM = WAVES * 16;
   constexpr int TN = FRAGS * 16;
   u64 rt = (M + TM - 1) / TM, ct = (N + TN - 1) / TN;
-  hipLaunchKernelGGL((zg<WAVES, KT, REORDER, FRAGS>), dim3((unsigned)(rt * ct)),
-                     dim3(WAVES * 64), 0, 0, A, B, C, M, N, K, ct);
+  hipLaunchKernelGGL((zg<WAVES, KT, REORDER, FRAGS, GROUPC>),
+                     dim3((unsigned)(rt * ct)), dim3(WAVES * 64), 0, 0, A, B,
+                     C, M, N, K, ct);
 }
 
 int main(int argc, char** argv) {
@@ -157,20 +171,11 @@ int main(int argc, char** argv) {
   (void)hipDeviceSynchronize();
 
   Variant variants[] = {
-      {"w4 k16 chain (v0)", launch_zg<4, 16, 0>},
-      {"w4 k16 reorder", launch_zg<4, 16, 1>},
-      {"w4 k32 chain", launch_zg<4, 32, 0>},
-      {"w4 k32 reorder", launch_zg<4, 32, 1>},
-      {"w8 k16 chain", launch_zg<8, 16, 0>},
-      {"w8 k16 reorder", launch_zg<8, 16, 1>},
-      {"w8 k32 reorder", launch_zg<8, 32, 1>},
-      {"w8 k16 f8 (128x128)", launch_zg<8, 16, 0, 8>},
-      {"w8 k8 f8 (128x128)", launch_zg<8, 8, 0, 8>},
-      {"w4 k16 f8 (64x128)", launch_zg<4, 16, 0, 8>},
-      {"w16 k16 f4 (256x64)", launch_zg<16, 16, 0, 4>},
-      {"w4 k16 gauss", launch_zg<4, 16, 2>},
-      {"w8 k16 gauss", launch_zg<8, 16, 2>},
-      {"w8 k16 f8 gauss", launch_zg<8, 16, 2, 8>},
+      {"w8 k16 (base)", launch_zg<8, 16, 0>},
+      {"w8 k16 gc8", launch_zg<8, 16, 0, 4, 8>},
+      {"w8 k16 gc16", launch_zg<8, 16, 0, 4, 16>},
+      {"w8 k16 gc32", launch_zg<8, 16, 0, 4, 32>},
+      {"w8 k16 f8 gc8", launch_zg<8, 16, 0, 8, 8>},
   };
   double flops = 8.0 * M * N * K;
   hipEvent_t e0, e1;
