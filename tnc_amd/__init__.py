@@ -25,3 +25,9 @@ from .circuit import Circuit, Permutor
 from .connectivity import ConnectivityLayout, connectivity_edges
 from .builders import random_circuit, sycamore_circuit
 from .partition import find_partitioning, partition_tensor_network
+from .repartition import (
+    CommunicationScheme,
+    NaivePartitioningModel,
+    balance_partitions,
+    compute_solution,
+)

@@ -121,11 +121,27 @@ class DistPlan:
 
 
 def make_plan(tn: CompositeTensor, nranks: int, trials: int = 16,
-              size_cap=None, seed: int = 0) -> DistPlan:
-    """Partition + per-partition paths + fan-in path, all deterministic."""
+              size_cap=None, seed: int = 0, sa_seconds: float = 0.0) -> DistPlan:
+    """Partition + per-partition paths + fan-in path, all deterministic.
+
+    sa_seconds > 0 refines the initial min-cut partitioning with the
+    reference's simulated-annealing repartitioner (simulated_annealing.rs
+    semantics; see repartition.py) before path finding. Note: SA wall-time
+    budgets make the plan timing-dependent — every rank must either use
+    sa_seconds=0 or receive the same refined partitioning (bench.py runs SA
+    on rank 0's inputs deterministically by seeding and step count)."""
     if nranks == 1:
         raise ValueError("use the single-GPU engine for one rank")
     partitioning = find_partitioning(tn, nranks, seed=seed)
+    if sa_seconds > 0:
+        import numpy as np
+
+        from .repartition import NaivePartitioningModel, balance_partitions
+
+        model = NaivePartitioningModel(tn, nranks)
+        partitioning, _ = balance_partitions(
+            model, partitioning, np.random.default_rng(seed), sa_seconds
+        )
     ptn = partition_tensor_network(tn, partitioning)
     result = RandomGreedy(trials, size_cap=size_cap).find_path(ptn)
     return DistPlan(ptn, result.replace_path(), nranks)
