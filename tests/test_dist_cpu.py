@@ -108,3 +108,24 @@ def test_dist_fanin_four_parts_two_unused_ranks():
     # plan for 4 ranks executed on 4 processes; exercises rank mapping with
     # the final partition pinned to rank 0 (communication.rs:89-115)
     _run(4, 4)
+
+
+def test_make_plan_rqc36_eight_way():
+    """The benchmark's 8-way plan builds deterministically and respects the
+    rank mapping contract (final partition -> rank 0)."""
+    from tnc_amd.dist import make_plan
+    from tnc_amd.fixtures import load_fixture
+
+    tn, _, _ = load_fixture("rqc36")
+    plan = make_plan(tn, 8, trials=4, size_cap=4.0e9)
+    assert plan.nparts == 8
+    assert plan.used_ranks <= 8
+    assert sorted(plan.part_rank.values()) == list(range(plan.used_ranks))
+    final_part = plan.path.toplevel[-1][0]
+    assert plan.part_rank[final_part] == 0
+    assert len(plan.path.toplevel) == plan.nparts - 1
+    assert plan.total_flops() > 0
+    # deterministic across "ranks"
+    plan2 = make_plan(tn, 8, trials=4, size_cap=4.0e9)
+    assert plan2.part_rank == plan.part_rank
+    assert plan2.path.toplevel == plan.path.toplevel

@@ -114,3 +114,56 @@ def test_greedy_nested_composites():
     assert 0 in result.ssa_path.nested
     replace = result.replace_path()
     validate_path(replace)
+
+
+def test_optimal_small():
+    """Optimal finds a cost <= greedy on small nets and a valid path."""
+    from tnc_amd import Optimal
+
+    tn = _setup_complex()
+    g = Greedy().find_path(tn)
+    o = Optimal().find_path(tn)
+    assert o.flops <= g.flops
+    validate_path(o.replace_path())
+    # greedy is already optimal on the pinned simple net
+    tn2 = _setup_simple()
+    assert Optimal().find_path(tn2).flops == 600.0
+
+
+def test_optimal_disconnected():
+    from tnc_amd import Optimal
+
+    bd = {0: 3, 1: 2, 2: 2}
+    tn = CompositeTensor(
+        [
+            LeafTensor.new_from_map([0], bd),
+            LeafTensor.new_from_map([1], bd),
+            LeafTensor.new_from_map([2], bd),
+        ]
+    )
+    r = Optimal().find_path(tn)
+    validate_path(r.replace_path())
+    assert len(r.ssa_path.toplevel) == 2
+
+
+def test_optimal_matches_oracle_result():
+    import numpy as np
+
+    from oracle import contract_network
+    from oracle.adapters import network_to_otensors
+    from tnc_amd import Optimal
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+
+    tn = random_circuit(5, 4, 0.5, 0.5, 11, ConnectivityLayout.LINE
+                        if hasattr(ConnectivityLayout, "LINE")
+                        else ConnectivityLayout.Line(5))
+    if len(tn.tensors) > 14:
+        import pytest
+
+        pytest.skip("too many tensors for optimal")
+    ref = contract_network(network_to_otensors(tn),
+                           Greedy().find_path(tn).replace_path())
+    out = contract_network(network_to_otensors(tn),
+                           Optimal().find_path(tn).replace_path())
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-12)

@@ -220,6 +220,98 @@ class RandomGreedy(_CotengrustLike):
         return best_path or []
 
 
+class Optimal(_CotengrustLike):
+    """OptMethod::Optimal (cotengrust.rs:77-79): exhaustive search for the
+    minimum-op-count contraction tree. Subset dynamic programming over the
+    3^n subset pairs; practical to ~16 leaves (the reference's optimal is
+    equally exponential)."""
+
+    MAX_LEAVES = 16
+
+    def _optimize_single(self, leaves, external):
+        n = len(leaves)
+        if n == 0:
+            return []
+        if n == 1:
+            return []
+        assert n <= self.MAX_LEAVES, (
+            f"Optimal supports at most {self.MAX_LEAVES} tensors, got {n}"
+        )
+        views = {1 << i: leaves[i] for i in range(n)}
+        best = {1 << i: (0.0, None) for i in range(n)}  # mask -> (cost, (l, r))
+        full = (1 << n) - 1
+        # iterate masks by popcount so sub-results exist
+        masks = sorted(range(1, full + 1), key=lambda m: bin(m).count("1"))
+        for m in masks:
+            if m in best:
+                continue
+            best_cost, best_split = math.inf, None
+            # enumerate proper submasks s of m with s < m^s to dedupe
+            s = (m - 1) & m
+            while s:
+                o = m ^ s
+                if s < o and s in best and o in best:
+                    vs, vo = views[s], views[o]
+                    cost = best[s][0] + best[o][0] + (vs | vo).size()
+                    if cost < best_cost:
+                        best_cost, best_split = cost, (s, o)
+                s = (s - 1) & m
+            if best_split is not None:
+                best[m] = (best_cost, best_split)
+                vs, vo = views[best_split[0]], views[best_split[1]]
+                views[m] = vs ^ vo
+        if full not in best:
+            # disconnected network: contract components optimally, then
+            # combine by outer products smallest-first (like greedy's tail)
+            comps = []
+            remaining = full
+            while remaining:
+                # grow a connected component from the lowest set bit
+                seed = remaining & (-remaining)
+                comp = seed
+                changed = True
+                while changed:
+                    changed = False
+                    r = remaining & ~comp
+                    b = r
+                    while b:
+                        bit = b & (-b)
+                        if (views[comp] & views[bit]).legs:
+                            comp |= bit
+                            changed = True
+                        b &= b - 1
+                comps.append(comp)
+                remaining &= ~comp
+            ssa = []
+            next_id = n
+            ids = []
+            for comp in comps:
+                cid, next_id = self._emit(comp, best, ssa, n, next_id)
+                ids.append((views[comp].size(), -cid, cid))
+            heapq.heapify(ids)
+            while len(ids) > 1:
+                _, _, a = heapq.heappop(ids)
+                _, _, b = heapq.heappop(ids)
+                ssa.append((a, b))
+                heapq.heappush(ids, (1.0, -next_id, next_id))
+                next_id += 1
+            return ssa
+        ssa = []
+        self._emit(full, best, ssa, n, n)
+        return ssa
+
+    def _emit(self, mask, best, ssa, n, next_id):
+        """Post-order emission of the tree under `mask`; returns
+        (ssa id of mask's result, next free id)."""
+        if bin(mask).count("1") == 1:
+            return mask.bit_length() - 1, next_id
+        l, r = best[mask][1]
+        lid, next_id = self._emit(l, best, ssa, n, next_id)
+        rid, next_id = self._emit(r, best, ssa, n, next_id)
+        ssa.append((lid, rid))
+        return next_id, next_id + 1
+
+
 def _ssa_op_cost(leaves, ssa_path):
     """(op count, peak size) of an SSA path over leaves — the op count is
     prod-of-union-dims per step (contraction_cost.rs:49-52), the peak is
