@@ -68,6 +68,23 @@ int tn_einsum_c128_dev(const uint64_t* out_labels, const uint64_t* out_shape,
                        const int64_t* b_strides, const void* b_dev,
                        size_t b_ndim, void* out_dev, void* stream);
 
+/* complex64 variants (the config-5 precision/throughput trade: f32 MFMA
+ * runs at ~2x the f64 matrix rate). Same contract, 8-byte elements. */
+int tn_einsum_c64(const uint64_t* out_labels, const uint64_t* out_shape,
+                  size_t out_ndim, const uint64_t* a_labels,
+                  const uint64_t* a_shape, const int64_t* a_strides,
+                  const void* a_data, size_t a_ndim, const uint64_t* b_labels,
+                  const uint64_t* b_shape, const int64_t* b_strides,
+                  const void* b_data, size_t b_ndim, void* out_data);
+
+int tn_einsum_c64_dev(const uint64_t* out_labels, const uint64_t* out_shape,
+                      size_t out_ndim, const uint64_t* a_labels,
+                      const uint64_t* a_shape, const int64_t* a_strides,
+                      const void* a_dev, size_t a_ndim,
+                      const uint64_t* b_labels, const uint64_t* b_shape,
+                      const int64_t* b_strides, const void* b_dev,
+                      size_t b_ndim, void* out_dev, void* stream);
+
 /* Select the HIP device used by subsequently created nets / einsum calls. */
 int tn_set_device(int device);
 
@@ -75,8 +92,12 @@ int tn_set_device(int device);
 
 typedef struct tn_net tn_net;
 
-/* Create an executor bound to `device`. */
+/* Create an executor bound to `device` (complex128 tensors). */
 tn_net* tn_net_create(int device);
+
+/* Like tn_net_create with an element type: dtype 0 = complex128,
+ * 1 = complex64 (leaf/result buffers are then 8-byte elements). */
+tn_net* tn_net_create2(int device, int dtype);
 
 /* Reserve a device arena of `bytes` for intermediates and packing
  * workspaces (one hipMalloc; first-fit, stream-ordered reuse). Without a

@@ -183,6 +183,8 @@ __global__ __launch_bounds__(256) void k2(const double2* A, const double2* B,
   }
 }
 
+static void f32_probe();
+
 static void check(const char* name, const std::vector<double2>& got,
                   const std::vector<double2>& ref, int M, int N) {
   double mx = 0; int bad = 0; int first = -1;
@@ -234,5 +236,43 @@ int main() {
     check("k2 (1-acc x4)", C, R, M, N);
     hipFree(dA); hipFree(dB); hipFree(dC);
   }
+  f32_probe();
   return 0;
+}
+
+// f32 16x16x4 D-layout probe: computes one MFMA with asymmetric operands
+// and reports which (lane, reg) -> row mapping matches the CPU result.
+typedef float v4f __attribute__((ext_vector_type(4)));
+__global__ void k_f32probe(float* D) {
+  int l = threadIdx.x;
+  float a = (float)((l % 16) * 8 + (l / 16));       // A[i][k] = i*8+k
+  float b = (float)((l / 16) * 100 + (l % 16) * 3); // B[k][j] = 100k+3j
+  v4f acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) D[l * 4 + r] = acc[r];
+}
+
+static void f32_probe() {
+  float* dD;
+  (void)hipMalloc(&dD, 64 * 4 * sizeof(float));
+  hipLaunchKernelGGL(k_f32probe, dim3(1), dim3(64), 0, 0, dD);
+  std::vector<float> D(256);
+  (void)hipMemcpy(D.data(), dD, 256 * 4, hipMemcpyDeviceToHost);
+  // CPU reference C[i][j] = sum_k A[i][k]*B[k][j]
+  float C[16][16];
+  for (int i = 0; i < 16; ++i)
+    for (int j = 0; j < 16; ++j) {
+      float s = 0;
+      for (int k = 0; k < 4; ++k) s += (i * 8 + k) * (100.f * k + 3.f * j);
+      C[i][j] = s;
+    }
+  int ok_a = 0, ok_b = 0;
+  for (int l = 0; l < 64; ++l)
+    for (int r = 0; r < 4; ++r) {
+      int col = l % 16;
+      if (D[l * 4 + r] == C[(l / 16) * 4 + r][col]) ++ok_a;  // row=(l/16)*4+r
+      if (D[l * 4 + r] == C[4 * r + l / 16][col]) ++ok_b;    // row=4r+l/16
+    }
+  printf("f32 16x16x4 D-map: (l/16)*4+r matches %d/256; 4r+l/16 matches %d/256\n",
+         ok_a, ok_b);
 }

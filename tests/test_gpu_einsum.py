@@ -175,3 +175,59 @@ def test_gemm_splitk():
     a_labels = [0, 1] + list(range(10, 22))
     b_labels = list(range(10, 22)) + [2, 3]
     run_case(a_labels, [2] * 14, b_labels, [2] * 14, rtol=1e-10)
+
+
+def _run_c64(a_labels, a_shape, b_labels, b_shape, seed=0, rtol=2e-3):
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(seed)
+    a = _rand(a_shape, rng).astype(np.complex64)
+    b = _rand(b_shape, rng).astype(np.complex64)
+    out_labels, _ = symmetric_difference(a_labels, list(a.shape),
+                                         b_labels, list(b.shape))
+    ref = oracle.contract_ndarrays(out_labels, a_labels,
+                                   a.astype(np.complex128), b_labels,
+                                   b.astype(np.complex128))
+    got = hiplib.einsum_c64(out_labels, a_labels, a, b_labels, b)
+    assert got.dtype == np.complex64
+    np.testing.assert_allclose(got, ref, rtol=rtol, atol=1e-4)
+
+
+def test_c64_smallk():
+    _run_c64([0, 1, 2, 3, 4, 5], [2] * 6, [9, 10, 2, 4], [2] * 4)
+
+
+def test_c64_gemm_mfma():
+    # f32 MFMA path, M=N=K=256 with packing
+    _run_c64([0, 1, 2, 3], [16, 16, 16, 16], [4, 5, 2, 3], [16, 16, 16, 16])
+
+
+def test_c64_gemm_splitk():
+    _run_c64([0, 1], [64, 8192], [1, 2], [8192, 64], rtol=5e-3)
+
+
+def test_c64_dot():
+    _run_c64(list(range(16)), [2] * 16, list(range(16)), [2] * 16, rtol=5e-3)
+
+
+def test_c64_network_engine():
+    """Full network in c64 vs the c128 oracle (config-5 precision trade)."""
+    from tnc_amd import Greedy
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.executor import ContractionEngine
+
+    tn = random_circuit(14, 8, 0.5, 0.5, 5, ConnectivityLayout.EAGLE)
+    replace = Greedy().find_path(tn).replace_path()
+    eng = ContractionEngine(tn, replace, dtype="c64")
+    try:
+        eng.contract()
+        legs, data = eng.result()
+        assert data.dtype == np.complex64
+    finally:
+        eng.close()
+    from oracle import contract_network
+    from oracle.adapters import network_to_otensors
+
+    ref = contract_network(network_to_otensors(tn), replace)
+    np.testing.assert_allclose(data, ref.data, rtol=1e-3, atol=1e-5)

@@ -103,11 +103,12 @@ __device__ __forceinline__ i64 gather1(const GatherMap& m, u64 p) {
 // kernels
 // ---------------------------------------------------------------------------
 
-template <bool P2>
-__global__ void k_einsum_smallk(const double2* __restrict__ A,
-                                const double2* __restrict__ B,
-                                double2* __restrict__ C, u64 nout,
+template <bool P2, typename CT>
+__global__ void k_einsum_smallk(const CT* __restrict__ A,
+                                const CT* __restrict__ B,
+                                CT* __restrict__ C, u64 nout,
                                 GatherMap omap, GatherMap kmap, int K) {
+  using RT = decltype(CT{}.x);
   __shared__ i64 koffA[TN_SMALLK];
   __shared__ i64 koffB[TN_SMALLK];
   if (threadIdx.x < (unsigned)K) {
@@ -121,43 +122,44 @@ __global__ void k_einsum_smallk(const double2* __restrict__ A,
        p += gridDim.x * (u64)blockDim.x) {
     i64 oa, ob;
     gather2<P2>(omap, p, oa, ob);
-    double re = 0.0, im = 0.0;
+    RT re = 0, im = 0;
     for (int k = 0; k < K; ++k) {
-      double2 a = A[oa + koffA[k]];
-      double2 b = B[ob + koffB[k]];
+      CT a = A[oa + koffA[k]];
+      CT b = B[ob + koffB[k]];
       re = fma(a.x, b.x, fma(-a.y, b.y, re));
       im = fma(a.x, b.y, fma(a.y, b.x, im));
     }
-    C[p] = make_double2(re, im);
+    C[p] = CT{re, im};
   }
 }
 
 // skinny shapes with K > TN_SMALLK: offsets computed inline per k.
-template <bool P2>
-__global__ void k_einsum_anyk(const double2* __restrict__ A,
-                              const double2* __restrict__ B,
-                              double2* __restrict__ C, u64 nout, GatherMap omap,
+template <bool P2, typename CT>
+__global__ void k_einsum_anyk(const CT* __restrict__ A,
+                              const CT* __restrict__ B,
+                              CT* __restrict__ C, u64 nout, GatherMap omap,
                               GatherMap kmap, u64 K) {
+  using RT = decltype(CT{}.x);
   for (u64 p = blockIdx.x * (u64)blockDim.x + threadIdx.x; p < nout;
        p += gridDim.x * (u64)blockDim.x) {
     i64 oa, ob;
     gather2<P2>(omap, p, oa, ob);
-    double re = 0.0, im = 0.0;
+    RT re = 0, im = 0;
     for (u64 k = 0; k < K; ++k) {
       i64 ka, kb;
       gather2<P2>(kmap, k, ka, kb);
-      double2 a = A[oa + ka];
-      double2 b = B[ob + kb];
+      CT a = A[oa + ka];
+      CT b = B[ob + kb];
       re = fma(a.x, b.x, fma(-a.y, b.y, re));
       im = fma(a.x, b.y, fma(a.y, b.x, im));
     }
-    C[p] = make_double2(re, im);
+    C[p] = CT{re, im};
   }
 }
 
-template <bool P2>
-__global__ void k_dot_partial(const double2* __restrict__ A,
-                              const double2* __restrict__ B,
+template <bool P2, typename CT>
+__global__ void k_dot_partial(const CT* __restrict__ A,
+                              const CT* __restrict__ B,
                               double2* __restrict__ ws, u64 K, GatherMap kmap) {
   __shared__ double sre[256], sim[256];
   double re = 0.0, im = 0.0;
@@ -165,10 +167,10 @@ __global__ void k_dot_partial(const double2* __restrict__ A,
        k += gridDim.x * (u64)blockDim.x) {
     i64 ka, kb;
     gather2<P2>(kmap, k, ka, kb);
-    double2 a = A[ka];
-    double2 b = B[kb];
-    re = fma(a.x, b.x, fma(-a.y, b.y, re));
-    im = fma(a.x, b.y, fma(a.y, b.x, im));
+    CT a = A[ka];
+    CT b = B[kb];
+    re = fma((double)a.x, (double)b.x, fma(-(double)a.y, (double)b.y, re));
+    im = fma((double)a.x, (double)b.y, fma((double)a.y, (double)b.x, im));
   }
   sre[threadIdx.x] = re;
   sim[threadIdx.x] = im;
@@ -183,8 +185,10 @@ __global__ void k_dot_partial(const double2* __restrict__ A,
   if (threadIdx.x == 0) ws[blockIdx.x] = make_double2(sre[0], sim[0]);
 }
 
-__global__ void k_dot_finish(const double2* __restrict__ ws, double2* out,
+template <typename CT>
+__global__ void k_dot_finish(const double2* __restrict__ ws, CT* out,
                              int nblocks) {
+  using RT = decltype(CT{}.x);
   __shared__ double sre[256], sim[256];
   double re = 0.0, im = 0.0;
   for (int i = threadIdx.x; i < nblocks; i += blockDim.x) {
@@ -201,15 +205,14 @@ __global__ void k_dot_finish(const double2* __restrict__ ws, double2* out,
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) *out = make_double2(sre[0], sim[0]);
+  if (threadIdx.x == 0) *out = CT{(RT)sre[0], (RT)sim[0]};
 }
 
 // dst[p] = src[gather(p)] — pack / unpack / general permute (also serves the
 // final-tensor Permutor, circuit_builder.rs:89-106, on device).
-template <bool P2>
-__global__ void k_permute_c128(const double2* __restrict__ src,
-                               double2* __restrict__ dst, u64 n,
-                               GatherMap map) {
+template <bool P2, typename CT>
+__global__ void k_permute_ct(const CT* __restrict__ src,
+                             CT* __restrict__ dst, u64 n, GatherMap map) {
   for (u64 p = blockIdx.x * (u64)blockDim.x + threadIdx.x; p < n;
        p += gridDim.x * (u64)blockDim.x) {
     dst[p] = src[gather1<P2>(map, p)];
@@ -223,12 +226,13 @@ __global__ void k_permute_c128(const double2* __restrict__ src,
 #define GT 64
 #define GK 16
 
+template <typename CT>
 __global__ __launch_bounds__(256) void k_zgemm_v1(
-    const double2* __restrict__ A, const double2* __restrict__ B,
-    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
+    const CT* __restrict__ A, const CT* __restrict__ B,
+    CT* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
     u64 kchunk) {
-  __shared__ double2 As[GT][GK + 1];
-  __shared__ double2 Bs[GK][GT + 1];
+  __shared__ CT As[GT][GK + 1];
+  __shared__ CT Bs[GK][GT + 1];
   const int tx = threadIdx.x % 16, ty = threadIdx.x / 16;
   const u64 tile = blockIdx.x % tiles;
   const u64 slice = blockIdx.x / tiles;
@@ -236,23 +240,23 @@ __global__ __launch_bounds__(256) void k_zgemm_v1(
   const u64 kbeg = slice * kchunk;
   const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
   C += slice * M * N;  // slice 0 == C itself when kchunk == K
-  double2 acc[4][4];
+  CT acc[4][4];
   for (int i = 0; i < 4; ++i)
-    for (int j = 0; j < 4; ++j) acc[i][j] = make_double2(0.0, 0.0);
+    for (int j = 0; j < 4; ++j) acc[i][j] = CT{0, 0};
   for (u64 k0 = kbeg; k0 < kend; k0 += GK) {
     for (int i = threadIdx.x; i < GT * GK; i += 256) {
       int r = i / GK, c = i % GK;
       As[r][c] = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
-                                                 : make_double2(0.0, 0.0);
+                                                 : CT{0, 0};
     }
     for (int i = threadIdx.x; i < GK * GT; i += 256) {
       int r = i / GT, c = i % GT;
       Bs[r][c] = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
-                                                 : make_double2(0.0, 0.0);
+                                                 : CT{0, 0};
     }
     __syncthreads();
     for (int kk = 0; kk < GK; ++kk) {
-      double2 a[4], b[4];
+      CT a[4], b[4];
       for (int i = 0; i < 4; ++i) a[i] = As[ty * 4 + i][kk];
       for (int j = 0; j < 4; ++j) b[j] = Bs[kk][tx * 4 + j];
       for (int i = 0; i < 4; ++i)
@@ -282,22 +286,53 @@ __global__ __launch_bounds__(256) void k_zgemm_v1(
 // 4 MFMAs per fragment (Cr += ArBr; Cr += (-Ai)Bi; Ci += ArBi; Ci += AiBr).
 // LDS staging is planar with padded rows to avoid bank conflicts.
 typedef double v4d __attribute__((ext_vector_type(4)));
+typedef float v4f __attribute__((ext_vector_type(4)));
+
+// Per-precision MFMA core. D-fragment row maps verified on hardware
+// (scripts/mfma_repro.hip): f64 16x16x4 -> row = 4*reg + lane/16,
+// f32 16x16x4 -> row = (lane/16)*4 + reg (they differ!).
+template <typename RT>
+struct MfmaCore;
+template <>
+struct MfmaCore<double> {
+  using acc_t = v4d;
+  static __device__ __forceinline__ acc_t mfma(double a, double b, acc_t c) {
+    return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+  }
+  static __device__ __forceinline__ int crow(int lane, int r) {
+    return 4 * r + lane / 16;
+  }
+};
+template <>
+struct MfmaCore<float> {
+  using acc_t = v4f;
+  static __device__ __forceinline__ acc_t mfma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+  static __device__ __forceinline__ int crow(int lane, int r) {
+    return (lane / 16) * 4 + r;
+  }
+};
 
 #define MF_T 128      // tile rows = MF_WAVES * 16
 #define MF_TN 64      // tile cols
 #define MF_K 16
 #define MF_THREADS 512
-#define A_LD 17  // padded row stride (doubles) for the A tiles
+#define A_LD 17  // padded row stride (elements) for the A tiles
 #define B_LD 66  // padded row stride for the B tiles
 
+template <typename CT>
 __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
-    const double2* __restrict__ A, const double2* __restrict__ B,
-    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
+    const CT* __restrict__ A, const CT* __restrict__ B,
+    CT* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
     u64 kchunk) {
-  __shared__ double Ar[MF_T * A_LD];
-  __shared__ double Ai[MF_T * A_LD];
-  __shared__ double Br[MF_K * B_LD];
-  __shared__ double Bi[MF_K * B_LD];
+  using RT = decltype(CT{}.x);
+  using Core = MfmaCore<RT>;
+  using acc_t = typename Core::acc_t;
+  __shared__ RT Ar[MF_T * A_LD];
+  __shared__ RT Ai[MF_T * A_LD];
+  __shared__ RT Br[MF_K * B_LD];
+  __shared__ RT Bi[MF_K * B_LD];
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
@@ -309,30 +344,29 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
   C += slice * M * N;
   const bool interior = (brow + MF_T <= M) && (bcol + MF_TN <= N);
 
-  v4d cr[4], ci[4];
+  acc_t cr[4], ci[4];
   for (int f = 0; f < 4; ++f) {
-    cr[f] = v4d{0.0, 0.0, 0.0, 0.0};
-    ci[f] = v4d{0.0, 0.0, 0.0, 0.0};
+    cr[f] = acc_t{0, 0, 0, 0};
+    ci[f] = acc_t{0, 0, 0, 0};
   }
 
-  // v_mfma_f64_16x16x4_f64 operand map (verified on hardware,
-  // scripts/mfma_repro.hip): lane l holds A[i = l%16][k = l/16] and
-  // B[k = l/16][j = l%16]; D: lane l, reg r -> row 4*r + l/16, col l%16.
+  // operand map (both precisions): lane l holds A[i = l%16][k = l/16] and
+  // B[k = l/16][j = l%16]; the D row map is per-precision (MfmaCore::crow).
   const int fi = lane % 16;
   const int fk = lane / 16;
 
   for (u64 k0 = kbeg; k0 < kend; k0 += MF_K) {
     for (int i = threadIdx.x; i < MF_T * MF_K; i += MF_THREADS) {
       int r = i / MF_K, c = i % MF_K;
-      double2 v = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
-                                                  : make_double2(0.0, 0.0);
+      CT v = (brow + r < M && k0 + c < kend) ? A[(brow + r) * K + k0 + c]
+                                             : CT{0, 0};
       Ar[r * A_LD + c] = v.x;
       Ai[r * A_LD + c] = v.y;
     }
     for (int i = threadIdx.x; i < MF_K * MF_TN; i += MF_THREADS) {
       int r = i / MF_TN, c = i % MF_TN;
-      double2 v = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
-                                                  : make_double2(0.0, 0.0);
+      CT v = (k0 + r < kend && bcol + c < N) ? B[(k0 + r) * N + bcol + c]
+                                             : CT{0, 0};
       Br[r * B_LD + c] = v.x;
       Bi[r * B_LD + c] = v.y;
     }
@@ -340,46 +374,46 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
     for (int kq = 0; kq < MF_K / 4; ++kq) {
       const int arow = wave * 16 + fi;
       const int ak = kq * 4 + fk;
-      const double ar = Ar[arow * A_LD + ak];
-      const double ai = Ai[arow * A_LD + ak];
+      const RT ar = Ar[arow * A_LD + ak];
+      const RT ai = Ai[arow * A_LD + ak];
       for (int f = 0; f < 4; ++f) {
         const int bcolf = f * 16 + fi;
-        const double br = Br[ak * B_LD + bcolf];
-        const double bi = Bi[ak * B_LD + bcolf];
-        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(ar, br, cr[f], 0, 0, 0);
-        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-ai, bi, cr[f], 0, 0, 0);
-        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(ar, bi, ci[f], 0, 0, 0);
-        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(ai, br, ci[f], 0, 0, 0);
+        const RT br = Br[ak * B_LD + bcolf];
+        const RT bi = Bi[ak * B_LD + bcolf];
+        cr[f] = Core::mfma(ar, br, cr[f]);
+        cr[f] = Core::mfma(-ai, bi, cr[f]);
+        ci[f] = Core::mfma(ar, bi, ci[f]);
+        ci[f] = Core::mfma(ai, br, ci[f]);
       }
     }
     __syncthreads();
   }
 
-  const int crow0 = wave * 16 + (lane / 16);
   const int ccol = lane % 16;
   for (int f = 0; f < 4; ++f) {
     for (int r = 0; r < 4; ++r) {
-      u64 row = brow + crow0 + 4 * r;
+      u64 row = brow + wave * 16 + Core::crow(lane, r);
       u64 col = bcol + f * 16 + ccol;
       if (interior || (row < M && col < N))
-        C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+        C[row * N + col] = CT{cr[f][r], ci[f][r]};
     }
   }
 }
 
 // split-K reduce: C[p] = sum over slices of ws[s][p]
-__global__ void k_splitk_reduce(const double2* __restrict__ ws,
-                                double2* __restrict__ C, u64 nout,
-                                int slices) {
+template <typename CT>
+__global__ void k_splitk_reduce(const CT* __restrict__ ws,
+                                CT* __restrict__ C, u64 nout, int slices) {
+  using RT = decltype(CT{}.x);
   for (u64 p = blockIdx.x * (u64)blockDim.x + threadIdx.x; p < nout;
        p += gridDim.x * (u64)blockDim.x) {
-    double re = 0.0, im = 0.0;
+    RT re = 0, im = 0;
     for (int s = 0; s < slices; ++s) {
-      double2 v = ws[(u64)s * nout + p];
+      CT v = ws[(u64)s * nout + p];
       re += v.x;
       im += v.y;
     }
-    C[p] = make_double2(re, im);
+    C[p] = CT{re, im};
   }
 }
 
@@ -392,7 +426,7 @@ struct Meta {
   u64 labels[TN_MAXR];
   u64 dims[TN_MAXR];
   i64 strides[TN_MAXR];  // in elements
-  const double2* data;
+  const void* data;
 };
 
 struct AxisInfo {
@@ -562,10 +596,13 @@ static int grid_for(u64 nout, int block = 256) {
 }
 
 // core einsum over device buffers; out is contiguous row-major in out order.
+template <typename CT>
 static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
                            int out_nd, const Meta& A, const Meta& B,
-                           double2* out, hipStream_t stream, WsCtx& ws,
+                           CT* out, hipStream_t stream, WsCtx& ws,
                            StepStats* stats) {
+  const CT* Adata = (const CT*)A.data;
+  const CT* Bdata = (const CT*)B.data;
   if (A.nd > TN_MAXR || B.nd > TN_MAXR || out_nd > TN_MAXR)
     FAILV(TN_ERR_INVALID, "tensor rank exceeds %d", TN_MAXR);
   auto find = [](const Meta& t, u64 lab) {
@@ -655,11 +692,11 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       if (rc_) return rc_;
     }
     if (kmap.pow2)
-      k_dot_partial<true><<<blocks, 256, 0, stream>>>(A.data, B.data, wsbuf,
-                                                      K, kmap);
+      k_dot_partial<true><<<blocks, 256, 0, stream>>>(Adata, Bdata, wsbuf, K,
+                                                      kmap);
     else
-      k_dot_partial<false><<<blocks, 256, 0, stream>>>(A.data, B.data, wsbuf,
-                                                       K, kmap);
+      k_dot_partial<false><<<blocks, 256, 0, stream>>>(Adata, Bdata, wsbuf, K,
+                                                       kmap);
     k_dot_finish<<<1, 256, 0, stream>>>(wsbuf, out, blocks);
     ws_free(ws, wsbuf);
     HIP_CHECK(hipGetLastError());
@@ -690,16 +727,16 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (K <= TN_SMALLK) {
       if (p2)
         k_einsum_smallk<true><<<blocks, 256, 0, stream>>>(
-            A.data, B.data, out, nout, omap, kmap, (int)K);
+            Adata, Bdata, out, nout, omap, kmap, (int)K);
       else
         k_einsum_smallk<false><<<blocks, 256, 0, stream>>>(
-            A.data, B.data, out, nout, omap, kmap, (int)K);
+            Adata, Bdata, out, nout, omap, kmap, (int)K);
     } else {
       if (p2)
-        k_einsum_anyk<true><<<blocks, 256, 0, stream>>>(A.data, B.data, out,
+        k_einsum_anyk<true><<<blocks, 256, 0, stream>>>(Adata, Bdata, out,
                                                         nout, omap, kmap, K);
       else
-        k_einsum_anyk<false><<<blocks, 256, 0, stream>>>(A.data, B.data, out,
+        k_einsum_anyk<false><<<blocks, 256, 0, stream>>>(Adata, Bdata, out,
                                                          nout, omap, kmap, K);
     }
     HIP_CHECK(hipGetLastError());
@@ -729,10 +766,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     return true;
   };
 
-  const double2* Ag = A.data;
-  const double2* Bg = B.data;
-  double2* packA = nullptr;
-  double2* packB = nullptr;
+  const CT* Ag = Adata;
+  const CT* Bg = Bdata;
+  CT* packA = nullptr;
+  CT* packB = nullptr;
   if (!is_ready(A, a_axes)) {
     std::vector<AxisInfo> ax;
     for (int axis : a_axes) ax.push_back({A.dims[axis], A.strides[axis], 0});
@@ -740,16 +777,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = M * K;
     {
-      int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(double2));
+      int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(CT));
       if (rc_) return rc_;
     }
     int blocks = grid_for(elems);
     if (map.pow2)
-      k_permute_c128<true><<<blocks, 256, 0, stream>>>(A.data, packA, elems,
-                                                       map);
+      k_permute_ct<true><<<blocks, 256, 0, stream>>>(Adata, packA, elems, map);
     else
-      k_permute_c128<false><<<blocks, 256, 0, stream>>>(A.data, packA, elems,
-                                                        map);
+      k_permute_ct<false><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
+                                                      map);
     Ag = packA;
   }
   if (!is_ready(B, b_axes)) {
@@ -759,16 +795,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = K * N;
     {
-      int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(double2));
+      int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(CT));
       if (rc_) return rc_;
     }
     int blocks = grid_for(elems);
     if (map.pow2)
-      k_permute_c128<true><<<blocks, 256, 0, stream>>>(B.data, packB, elems,
-                                                       map);
+      k_permute_ct<true><<<blocks, 256, 0, stream>>>(Bdata, packB, elems, map);
     else
-      k_permute_c128<false><<<blocks, 256, 0, stream>>>(B.data, packB, elems,
-                                                        map);
+      k_permute_ct<false><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
+                                                      map);
     Bg = packB;
   }
 
@@ -776,12 +811,12 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   bool direct = true;
   if (!m_out.empty() && !n_out.empty() && m_out.back() > n_out.front())
     direct = false;
-  double2* Cg = out;
-  double2* tmpC = nullptr;
+  CT* Cg = out;
+  CT* tmpC = nullptr;
   if (!direct) {
     if (stats) stats->kind = 3;
     {
-      int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(double2));
+      int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(CT));
       if (rc_) return rc_;
     }
     Cg = tmpC;
@@ -803,10 +838,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   u64 kchunk = (K + splitk - 1) / splitk;
   kchunk = ((kchunk + MF_K - 1) / MF_K) * MF_K;  // tile-aligned
   splitk = (K + kchunk - 1) / kchunk;
-  double2* gemm_out = Cg;
-  double2* splitbuf = nullptr;
+  CT* gemm_out = Cg;
+  CT* splitbuf = nullptr;
   if (splitk > 1) {
-    int rc_ = ws_alloc(ws, (void**)&splitbuf, splitk * nout * sizeof(double2));
+    int rc_ = ws_alloc(ws, (void**)&splitbuf, splitk * nout * sizeof(CT));
     if (rc_) return rc_;
     gemm_out = splitbuf;
   }
@@ -847,9 +882,11 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     int blocks = grid_for(nout);
     if (map.pow2)
-      k_permute_c128<true><<<blocks, 256, 0, stream>>>(tmpC, out, nout, map);
+      k_permute_ct<true><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
+                                                     nout, map);
     else
-      k_permute_c128<false><<<blocks, 256, 0, stream>>>(tmpC, out, nout, map);
+      k_permute_ct<false><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
+                                                      nout, map);
     HIP_CHECK(hipGetLastError());
   }
   ws_free(ws, packA);
@@ -899,13 +936,14 @@ static int fill_meta(Meta* m, const u64* labels, const u64* shape,
   return TN_OK;
 }
 
-extern "C" int tn_einsum_c128_dev(const u64* out_labels, const u64* out_shape,
-                                  size_t out_ndim, const u64* a_labels,
-                                  const u64* a_shape, const i64* a_strides,
-                                  const void* a_dev, size_t a_ndim,
-                                  const u64* b_labels, const u64* b_shape,
-                                  const i64* b_strides, const void* b_dev,
-                                  size_t b_ndim, void* out_dev, void* stream) {
+template <typename CT>
+static int einsum_dev_entry(const u64* out_labels, const u64* out_shape,
+                            size_t out_ndim, const u64* a_labels,
+                            const u64* a_shape, const i64* a_strides,
+                            const void* a_dev, size_t a_ndim,
+                            const u64* b_labels, const u64* b_shape,
+                            const i64* b_strides, const void* b_dev,
+                            size_t b_ndim, void* out_dev, void* stream) {
   int rc = require_gpu();
   if (rc) return rc;
   Meta A, B;
@@ -914,8 +952,34 @@ extern "C" int tn_einsum_c128_dev(const u64* out_labels, const u64* out_shape,
   rc = fill_meta(&B, b_labels, b_shape, b_strides, b_dev, b_ndim);
   if (rc) return rc;
   WsCtx ws{nullptr, (hipStream_t)stream};
-  return einsum_dev_impl(out_labels, out_shape, (int)out_ndim, A, B,
-                         (double2*)out_dev, (hipStream_t)stream, ws, nullptr);
+  return einsum_dev_impl<CT>(out_labels, out_shape, (int)out_ndim, A, B,
+                             (CT*)out_dev, (hipStream_t)stream, ws, nullptr);
+}
+
+extern "C" int tn_einsum_c128_dev(const u64* out_labels, const u64* out_shape,
+                                  size_t out_ndim, const u64* a_labels,
+                                  const u64* a_shape, const i64* a_strides,
+                                  const void* a_dev, size_t a_ndim,
+                                  const u64* b_labels, const u64* b_shape,
+                                  const i64* b_strides, const void* b_dev,
+                                  size_t b_ndim, void* out_dev, void* stream) {
+  return einsum_dev_entry<double2>(out_labels, out_shape, out_ndim, a_labels,
+                                   a_shape, a_strides, a_dev, a_ndim, b_labels,
+                                   b_shape, b_strides, b_dev, b_ndim, out_dev,
+                                   stream);
+}
+
+extern "C" int tn_einsum_c64_dev(const u64* out_labels, const u64* out_shape,
+                                 size_t out_ndim, const u64* a_labels,
+                                 const u64* a_shape, const i64* a_strides,
+                                 const void* a_dev, size_t a_ndim,
+                                 const u64* b_labels, const u64* b_shape,
+                                 const i64* b_strides, const void* b_dev,
+                                 size_t b_ndim, void* out_dev, void* stream) {
+  return einsum_dev_entry<float2>(out_labels, out_shape, out_ndim, a_labels,
+                                  a_shape, a_strides, a_dev, a_ndim, b_labels,
+                                  b_shape, b_strides, b_dev, b_ndim, out_dev,
+                                  stream);
 }
 
 static u64 span_elems(const u64* shape, const i64* strides, size_t nd) {
@@ -930,17 +994,19 @@ static u64 span_elems(const u64* shape, const i64* strides, size_t nd) {
   return span;
 }
 
-extern "C" int tn_einsum_c128(const u64* out_labels, const u64* out_shape,
-                              size_t out_ndim, const u64* a_labels,
-                              const u64* a_shape, const i64* a_strides,
-                              const void* a_data, size_t a_ndim,
-                              const u64* b_labels, const u64* b_shape,
-                              const i64* b_strides, const void* b_data,
-                              size_t b_ndim, void* out_data) {
+template <typename CT>
+static int einsum_host_entry(const u64* out_labels, const u64* out_shape,
+                             size_t out_ndim, const u64* a_labels,
+                             const u64* a_shape, const i64* a_strides,
+                             const void* a_data, size_t a_ndim,
+                             const u64* b_labels, const u64* b_shape,
+                             const i64* b_strides, const void* b_data,
+                             size_t b_ndim, void* out_data) {
   int rc = require_gpu();
   if (rc) return rc;
   HIP_CHECK(hipSetDevice(g_device));
   ensure_mempool(g_device);
+  const size_t es = sizeof(CT);
   if (a_strides)
     for (size_t i = 0; i < a_ndim; ++i)
       if (a_strides[i] < 0) FAILV(TN_ERR_INVALID, "negative strides");
@@ -952,21 +1018,48 @@ extern "C" int tn_einsum_c128(const u64* out_labels, const u64* out_shape,
   u64 out_elems = 1;
   for (size_t i = 0; i < out_ndim; ++i) out_elems *= out_shape[i];
   void *da, *db, *dout;
-  HIP_CHECK(hipMalloc(&da, a_span * 16));
-  HIP_CHECK(hipMalloc(&db, b_span * 16));
-  HIP_CHECK(hipMalloc(&dout, out_elems * 16));
-  HIP_CHECK(hipMemcpy(da, a_data, a_span * 16, hipMemcpyHostToDevice));
-  HIP_CHECK(hipMemcpy(db, b_data, b_span * 16, hipMemcpyHostToDevice));
-  rc = tn_einsum_c128_dev(out_labels, out_shape, out_ndim, a_labels, a_shape,
-                          a_strides, da, a_ndim, b_labels, b_shape, b_strides,
-                          db, b_ndim, dout, nullptr);
+  HIP_CHECK(hipMalloc(&da, a_span * es));
+  HIP_CHECK(hipMalloc(&db, b_span * es));
+  HIP_CHECK(hipMalloc(&dout, out_elems * es));
+  HIP_CHECK(hipMemcpy(da, a_data, a_span * es, hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(db, b_data, b_span * es, hipMemcpyHostToDevice));
+  rc = einsum_dev_entry<CT>(out_labels, out_shape, out_ndim, a_labels,
+                            a_shape, a_strides, da, a_ndim, b_labels, b_shape,
+                            b_strides, db, b_ndim, dout, nullptr);
   if (rc == TN_OK) {
-    HIP_CHECK(hipMemcpy(out_data, dout, out_elems * 16, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(out_data, dout, out_elems * es,
+                        hipMemcpyDeviceToHost));
   }
   (void)hipFree(da);
   (void)hipFree(db);
   (void)hipFree(dout);
   return rc;
+}
+
+extern "C" int tn_einsum_c128(const u64* out_labels, const u64* out_shape,
+                              size_t out_ndim, const u64* a_labels,
+                              const u64* a_shape, const i64* a_strides,
+                              const void* a_data, size_t a_ndim,
+                              const u64* b_labels, const u64* b_shape,
+                              const i64* b_strides, const void* b_data,
+                              size_t b_ndim, void* out_data) {
+  return einsum_host_entry<double2>(out_labels, out_shape, out_ndim, a_labels,
+                                    a_shape, a_strides, a_data, a_ndim,
+                                    b_labels, b_shape, b_strides, b_data,
+                                    b_ndim, out_data);
+}
+
+extern "C" int tn_einsum_c64(const u64* out_labels, const u64* out_shape,
+                             size_t out_ndim, const u64* a_labels,
+                             const u64* a_shape, const i64* a_strides,
+                             const void* a_data, size_t a_ndim,
+                             const u64* b_labels, const u64* b_shape,
+                             const i64* b_strides, const void* b_data,
+                             size_t b_ndim, void* out_data) {
+  return einsum_host_entry<float2>(out_labels, out_shape, out_ndim, a_labels,
+                                   a_shape, a_strides, a_data, a_ndim,
+                                   b_labels, b_shape, b_strides, b_data,
+                                   b_ndim, out_data);
 }
 
 // ---------------------------------------------------------------------------
@@ -976,7 +1069,7 @@ extern "C" int tn_einsum_c128(const u64* out_labels, const u64* out_shape,
 struct DevTensor {
   std::vector<u64> labels;
   std::vector<u64> dims;
-  double2* data = nullptr;
+  void* data = nullptr;
   u64 elems = 1;
   bool owned = false;     // intermediate owned by the walk (freed on consume)
   bool external = false;  // caller-owned device buffer (never freed by us)
@@ -984,6 +1077,8 @@ struct DevTensor {
 
 struct tn_net {
   int device = 0;
+  int dtype = 0;   // 0 = c128, 1 = c64
+  size_t esize = 16;
   hipStream_t stream = nullptr;
   std::vector<DevTensor> leaves;
   DevTensor final_t;       // final tensor of the last contract (owned)
@@ -993,7 +1088,11 @@ struct tn_net {
   u64 pool_in_use = 0;
 };
 
-extern "C" tn_net* tn_net_create(int device) {
+extern "C" tn_net* tn_net_create2(int device, int dtype) {
+  if (dtype != 0 && dtype != 1) {
+    g_last_error = "dtype must be 0 (c128) or 1 (c64)";
+    return nullptr;
+  }
   if (tn_device_count() == 0) {
     g_last_error = "no AMD GPU present — tnc_hip has no CPU fallback by design";
     return nullptr;
@@ -1005,12 +1104,18 @@ extern "C" tn_net* tn_net_create(int device) {
   ensure_mempool(device);
   tn_net* net = new tn_net();
   net->device = device;
+  net->dtype = dtype;
+  net->esize = dtype == 0 ? 16 : 8;
   if (hipStreamCreate(&net->stream) != hipSuccess) {
     delete net;
     g_last_error = "hipStreamCreate failed";
     return nullptr;
   }
   return net;
+}
+
+extern "C" tn_net* tn_net_create(int device) {
+  return tn_net_create2(device, 0);
 }
 
 extern "C" int tn_net_reserve(tn_net* net, uint64_t bytes) {
@@ -1035,12 +1140,12 @@ extern "C" int64_t tn_net_add_leaf(tn_net* net, const u64* labels,
   t.labels.assign(labels, labels + ndim);
   t.dims.assign(dims, dims + ndim);
   for (size_t i = 0; i < ndim; ++i) t.elems *= dims[i];
-  if (hipMalloc((void**)&t.data, t.elems * 16) != hipSuccess) {
+  if (hipMalloc((void**)&t.data, t.elems * net->esize) != hipSuccess) {
     g_last_error = "hipMalloc failed for leaf";
     return -TN_ERR_OOM;
   }
-  if (hipMemcpy(t.data, host_data, t.elems * 16, hipMemcpyHostToDevice) !=
-      hipSuccess) {
+  if (hipMemcpy(t.data, host_data, t.elems * net->esize,
+                hipMemcpyHostToDevice) != hipSuccess) {
     (void)hipFree(t.data);
     g_last_error = "hipMemcpy failed for leaf";
     return -TN_ERR_HIP;
@@ -1062,7 +1167,7 @@ extern "C" int64_t tn_net_add_leaf_dev(tn_net* net, const u64* labels,
   t.labels.assign(labels, labels + ndim);
   t.dims.assign(dims, dims + ndim);
   for (size_t i = 0; i < ndim; ++i) t.elems *= dims[i];
-  t.data = (double2*)dev_data;
+  t.data = dev_data;
   t.owned = false;
   t.external = true;
   net->leaves.push_back(std::move(t));
@@ -1143,7 +1248,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     symdiff(A, B, &out.labels, &out.dims);
     out.elems = 1;
     for (u64 d : out.dims) out.elems *= d;
-    if (ws_alloc(ws, (void**)&out.data, out.elems * 16) != TN_OK) {
+    if (ws_alloc(ws, (void**)&out.data, out.elems * net->esize) != TN_OK) {
       rc = TN_ERR_OOM;
       g_last_error = "device allocation failed for intermediate";
       break;
@@ -1179,9 +1284,14 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
       st.gemm_ev0 = ev[4 * s + 2];
       st.gemm_ev1 = ev[4 * s + 3];
     }
-    rc = einsum_dev_impl(out.labels.data(), out.dims.data(),
-                         (int)out.labels.size(), ma, mb, out.data, net->stream,
-                         ws, &st);
+    if (net->dtype == 0)
+      rc = einsum_dev_impl<double2>(out.labels.data(), out.dims.data(),
+                                    (int)out.labels.size(), ma, mb,
+                                    (double2*)out.data, net->stream, ws, &st);
+    else
+      rc = einsum_dev_impl<float2>(out.labels.data(), out.dims.data(),
+                                   (int)out.labels.size(), ma, mb,
+                                   (float2*)out.data, net->stream, ws, &st);
     if (profiled) HIP_CHECK(hipEventRecord(ev[4 * s + 1], net->stream));
     if (kind) kind[s] = st.kind;
     if (profiled) kinds_local[s] = st.kind;
@@ -1234,11 +1344,11 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
       if (!f.owned) {
         // final == a leaf (empty path): copy so result is stable
         DevTensor c = f;
-        if (hipMalloc((void**)&c.data, f.elems * 16) != hipSuccess) {
+        if (hipMalloc((void**)&c.data, f.elems * net->esize) != hipSuccess) {
           rc = TN_ERR_OOM;
           g_last_error = "hipMalloc failed for final copy";
         } else {
-          HIP_CHECK(hipMemcpy(c.data, f.data, f.elems * 16,
+          HIP_CHECK(hipMemcpy(c.data, f.data, f.elems * net->esize,
                               hipMemcpyDeviceToDevice));
           c.owned = true;
           net->final_t = std::move(c);
@@ -1297,7 +1407,8 @@ extern "C" int tn_net_result_meta(tn_net* net, u64* labels, u64* dims,
 extern "C" int tn_net_result_data(tn_net* net, void* host_out) {
   if (!net || !net->has_final) FAILV(TN_ERR_INVALID, "no result available");
   HIP_CHECK(hipSetDevice(net->device));
-  HIP_CHECK(hipMemcpy(host_out, net->final_t.data, net->final_t.elems * 16,
+  HIP_CHECK(hipMemcpy(host_out, net->final_t.data,
+                      net->final_t.elems * net->esize,
                       hipMemcpyDeviceToHost));
   return TN_OK;
 }

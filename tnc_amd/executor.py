@@ -48,16 +48,15 @@ def plan_steps(leaves, steps):
     return infos
 
 
-def arena_bytes(leaves, steps, infos):
+def arena_bytes(leaves, steps, infos, esize=16):
     """Peak device-arena demand: live intermediates + this step's output and
     worst-case packing workspaces (A' + B' + unpacked C), walked over the
-    plan. 16 B per c128 element; padded 15% for first-fit fragmentation."""
-    nleaves = len(leaves)
+    plan. `esize` bytes per element; padded 15% for fragmentation."""
     live = {}  # slot -> bytes (intermediates only; leaves live outside)
     peak = 0
     for info in infos:
-        out_b = info.m * info.n * 16
-        extra = (info.m * info.k + info.k * info.n) * 16 + out_b  # packs+tmpC
+        out_b = info.m * info.n * esize
+        extra = (info.m * info.k + info.k * info.n) * esize + out_b
         demand = sum(live.values()) + out_b + extra
         peak = max(peak, demand)
         live.pop(info.j, None)
@@ -68,9 +67,14 @@ def arena_bytes(leaves, steps, infos):
 class ContractionEngine:
     """Device-resident executor for one (flattened) network."""
 
-    def __init__(self, tn: CompositeTensor, replace_path: ContractionPath, device=0):
+    def __init__(self, tn: CompositeTensor, replace_path: ContractionPath,
+                 device=0, dtype="c128"):
         if hiplib.device_count() == 0:
             raise RuntimeError("no AMD GPU present — tnc_amd has no CPU fallback")
+        assert dtype in ("c128", "c64")
+        self.dtype = dtype
+        self.npdtype = np.complex128 if dtype == "c128" else np.complex64
+        self.esize = 16 if dtype == "c128" else 8
         leaves, steps, final = flatten_network(tn, replace_path)
         self.leaves = leaves
         self.steps = steps
@@ -79,10 +83,10 @@ class ContractionEngine:
         self.total_flops = sum(s.flops for s in self.infos)
         L = hiplib.lib()
         hiplib.check(L.tn_set_device(device), "tn_set_device")
-        self.net = L.tn_net_create(device)
+        self.net = L.tn_net_create2(device, 0 if dtype == "c128" else 1)
         if not self.net:
             raise RuntimeError(f"tn_net_create failed: {hiplib.last_error()}")
-        reserve = arena_bytes(leaves, steps, self.infos)
+        reserve = arena_bytes(leaves, steps, self.infos, self.esize)
         # arena is an optimization: if the reservation fails (tiny GPUs,
         # fragmented memory), the executor falls back to hipMallocAsync
         if reserve > 64 * 1024 * 1024:
@@ -95,7 +99,7 @@ class ContractionEngine:
                     f"({hiplib.last_error()}); falling back to async allocs"
                 )
         for t in leaves:
-            data = np.ascontiguousarray(t.tensordata.into_data(), dtype=np.complex128)
+            data = np.ascontiguousarray(t.tensordata.into_data(), dtype=self.npdtype)
             assert list(data.shape) == list(t.bond_dims), (data.shape, t.bond_dims)
             idx = L.tn_net_add_leaf(
                 self.net,
@@ -146,7 +150,7 @@ class ContractionEngine:
         )
         shape = tuple(dims[i] for i in range(nd.value))
         legs = [labels[i] for i in range(nd.value)]
-        out = np.empty(shape, dtype=np.complex128)
+        out = np.empty(shape, dtype=self.npdtype)
         hiplib.check(
             L.tn_net_result_data(self.net, out.ctypes.data_as(ctypes.c_void_p)),
             "tn_net_result_data",

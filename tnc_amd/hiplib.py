@@ -45,6 +45,12 @@ def lib() -> ctypes.CDLL:
             u64p, u64p, i64p, ctypes.c_void_p, ctypes.c_size_t,
             ctypes.c_void_p, ctypes.c_void_p,
         ]
+        L.tn_einsum_c64.restype = ctypes.c_int
+        L.tn_einsum_c64.argtypes = L.tn_einsum_c128.argtypes
+        L.tn_einsum_c64_dev.restype = ctypes.c_int
+        L.tn_einsum_c64_dev.argtypes = L.tn_einsum_c128_dev.argtypes
+        L.tn_net_create2.restype = ctypes.c_void_p
+        L.tn_net_create2.argtypes = [ctypes.c_int, ctypes.c_int]
         L.tn_net_create.restype = ctypes.c_void_p
         L.tn_net_create.argtypes = [ctypes.c_int]
         L.tn_net_reserve.restype = ctypes.c_int
@@ -110,13 +116,10 @@ def _i64arr(vals):
     return (ctypes.c_int64 * len(vals))(*[int(v) for v in vals])
 
 
-def einsum_c128(out_labels, a_labels, a: np.ndarray, b_labels, b: np.ndarray,
-                out_shape=None) -> np.ndarray:
-    """Host-buffer einsum via the GPU (tn_einsum_c128): the direct
-    tblis::tensor_mult parity entry point. Accepts non-contiguous views
-    (strides forwarded in elements)."""
-    a = np.asarray(a, dtype=np.complex128)
-    b = np.asarray(b, dtype=np.complex128)
+def _einsum_host(out_labels, a_labels, a, b_labels, b, out_shape, npdtype,
+                 esize, fn_name):
+    a = np.asarray(a, dtype=npdtype)
+    b = np.asarray(b, dtype=npdtype)
     if out_shape is None:
         dimmap = {}
         for lab, d in zip(a_labels, a.shape):
@@ -124,10 +127,10 @@ def einsum_c128(out_labels, a_labels, a: np.ndarray, b_labels, b: np.ndarray,
         for lab, d in zip(b_labels, b.shape):
             dimmap[lab] = d
         out_shape = [dimmap[l] for l in out_labels]
-    out = np.empty(tuple(out_shape), dtype=np.complex128)
-    a_str = [s // 16 for s in a.strides]
-    b_str = [s // 16 for s in b.strides]
-    rc = lib().tn_einsum_c128(
+    out = np.empty(tuple(out_shape), dtype=npdtype)
+    a_str = [s // esize for s in a.strides]
+    b_str = [s // esize for s in b.strides]
+    rc = getattr(lib(), fn_name)(
         _u64arr(out_labels), _u64arr(out_shape), len(out_labels),
         _u64arr(a_labels), _u64arr(a.shape), _i64arr(a_str),
         a.ctypes.data_as(ctypes.c_void_p), a.ndim,
@@ -135,5 +138,21 @@ def einsum_c128(out_labels, a_labels, a: np.ndarray, b_labels, b: np.ndarray,
         b.ctypes.data_as(ctypes.c_void_p), b.ndim,
         out.ctypes.data_as(ctypes.c_void_p),
     )
-    check(rc, "tn_einsum_c128")
+    check(rc, fn_name)
     return out
+
+
+def einsum_c128(out_labels, a_labels, a: np.ndarray, b_labels, b: np.ndarray,
+                out_shape=None) -> np.ndarray:
+    """Host-buffer einsum via the GPU (tn_einsum_c128): the direct
+    tblis::tensor_mult parity entry point. Accepts non-contiguous views
+    (strides forwarded in elements)."""
+    return _einsum_host(out_labels, a_labels, a, b_labels, b, out_shape,
+                        np.complex128, 16, "tn_einsum_c128")
+
+
+def einsum_c64(out_labels, a_labels, a: np.ndarray, b_labels, b: np.ndarray,
+               out_shape=None) -> np.ndarray:
+    """complex64 einsum via the GPU (tn_einsum_c64, f32 MFMA path)."""
+    return _einsum_host(out_labels, a_labels, a, b_labels, b, out_shape,
+                        np.complex64, 8, "tn_einsum_c64")
