@@ -27,6 +27,7 @@ sys.path.insert(0, ROOT)
 # MI355X peaks (MI355X_MICROARCH.md; f64 matrix peak cross-checked by the
 # committed muBench under profiles/ — see DESIGN.md "Measurement")
 F64_MFMA_PEAK = 78.6e12   # real flops/s, v_mfma_f64_16x16x4_f64 dense
+F32_MFMA_PEAK = 157.3e12  # real flops/s, v_mfma_f32_16x16x4_f32 (c64 path)
 HBM_PEAK = 8.0e12         # bytes/s (spec; ~6.3 TB/s achievable)
 
 
@@ -42,13 +43,13 @@ def parse_args():
     return p.parse_args()
 
 
-def step_bytes(info):
+def step_bytes(info, esize=16):
     """Algorithmic bytes of one einsum step: read A and B once, write out
-    once (16 B per c128 element)."""
-    return 16.0 * (info.m * info.k + info.k * info.n + info.m * info.n)
+    once (esize bytes per element)."""
+    return float(esize) * (info.m * info.k + info.k * info.n + info.m * info.n)
 
 
-def roofline_from_profile(infos, step_ms, gemm_ms, kinds):
+def roofline_from_profile(infos, step_ms, gemm_ms, kinds, dtype="c128"):
     """Dominant kernel + its roofline leg from live HIP-event timings."""
     dom = max(range(len(step_ms)), key=lambda s: step_ms[s])
     info = infos[dom]
@@ -61,22 +62,24 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds):
             traffic = pmc.get("dominant_kernel_traffic_bytes")
         except Exception:
             traffic = None
+    mfma_peak = F64_MFMA_PEAK if dtype == "c128" else F32_MFMA_PEAK
+    esize = 16 if dtype == "c128" else 8
     if kinds[dom] >= 2 and gemm_ms[dom] > 0:
         dur_s = gemm_ms[dom] / 1e3
         achieved = info.flops / dur_s
         return {
             "bound": "mfma",
             "achieved": achieved,
-            "peak": F64_MFMA_PEAK,
+            "peak": mfma_peak,
             "unit": "FLOP/s",
-            "frac": achieved / F64_MFMA_PEAK,
+            "frac": achieved / mfma_peak,
             "traffic": traffic,
             "kernel": "k_zgemm_mfma",
             "launch_ms": gemm_ms[dom],
             "mnk": [info.m, info.n, info.k],
         }
     dur_s = step_ms[dom] / 1e3
-    achieved = step_bytes(info) / dur_s
+    achieved = step_bytes(info, esize) / dur_s
     return {
         "bound": "hbm",
         "achieved": achieved,
@@ -90,7 +93,8 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds):
     }
 
 
-def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29):
+def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29,
+                 dtype="c128"):
     """Oracle (numpy einsum -> BLAS zgemm) timed on the host cores over a
     bounded sample: the largest path steps whose operands fit `cap_elems`,
     random-valued inputs of the same shapes (einsum time is value-
@@ -112,10 +116,13 @@ def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29):
             continue
         # synthetic operands shaped like the step ([M legs][K legs] etc. as
         # flat 2-leg tensors: zgemm time depends on (M, N, K) only)
+        npdtype = np.complex128 if dtype == "c128" else np.complex64
         a = (rng.standard_normal((int(info.m), int(info.k)))
-             + 1j * rng.standard_normal((int(info.m), int(info.k))))
+             + 1j * rng.standard_normal((int(info.m), int(info.k)))
+             ).astype(npdtype)
         b = (rng.standard_normal((int(info.k), int(info.n)))
-             + 1j * rng.standard_normal((int(info.k), int(info.n))))
+             + 1j * rng.standard_normal((int(info.k), int(info.n)))
+             ).astype(npdtype)
         t0 = time.perf_counter()
         oracle.contract_ndarrays([0, 2], [0, 1], a, [1, 2], b)
         dt = time.perf_counter() - t0
@@ -137,7 +144,8 @@ def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29):
         "kind": "port",
         "sample": f"{used} largest {fixture} path steps with operands <= "
                   f"{cap_elems} elems ({100*total_flops/all_flops:.1f}% of "
-                  f"path flops), random-valued same-shape zgemm, "
+                  f"path flops), random-valued same-shape "
+                  f"{'zgemm' if dtype == 'c128' else 'cgemm'}, "
                   f"{total_time:.1f}s",
     }
 
@@ -152,13 +160,14 @@ def run_single(args):
     from tnc_amd.fixtures import load_fixture
 
     tn, replace_toplevel, meta = load_fixture(args.fixture)
+    dtype = meta.get("dtype", "c128")
     replace = ContractionPath.simple(replace_toplevel)
-    eng = ContractionEngine(tn, replace, device=0)
+    eng = ContractionEngine(tn, replace, device=0, dtype=dtype)
     flops_per_contraction = eng.total_flops
 
     # one profiled pass (doubles as extra warmup)
     _, step_ms, gemm_ms, kinds = eng.contract_profiled()
-    roofline = roofline_from_profile(eng.infos, step_ms, gemm_ms, kinds)
+    roofline = roofline_from_profile(eng.infos, step_ms, gemm_ms, kinds, dtype)
 
     for _ in range(args.warmup):
         eng.contract()
@@ -168,10 +177,11 @@ def run_single(args):
     wall = time.perf_counter() - t0
 
     value = flops_per_contraction * args.steps / wall / 1e9
-    cb = None if args.no_cpu_baseline else cpu_baseline(eng.infos, args.fixture)
+    cb = None if args.no_cpu_baseline else cpu_baseline(
+        eng.infos, args.fixture, dtype=dtype)
     eng.close()
     emit({
-        "metric": "pairwise-contraction GFLOP/s (c128)",
+        "metric": f"pairwise-contraction GFLOP/s ({dtype})",
         "value": value,
         "unit": "GFLOP/s",
         "n_gpus": 1,
@@ -181,7 +191,7 @@ def run_single(args):
         "higher_is_better": True,
         "scaling": "strong",
         "vs_baseline": None,
-        "dtype": "c128",
+        "dtype": dtype,
         "data": "synthetic",
         "config": {
             "workload": f"{args.fixture}: 36q depth-14 RQC single-amplitude "

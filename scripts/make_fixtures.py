@@ -34,6 +34,7 @@ def main():
         flops = metric_flops(tn, replace.toplevel)
         meta = {
             "config": {k: str(v) for k, v in cfg.items()},
+            "dtype": cfg.get("dtype", "c128"),
             "op_cost": rg.flops,
             "peak_size_elems": rg.size,
             "metric_flops": flops,

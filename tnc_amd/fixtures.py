@@ -12,7 +12,7 @@ import os
 
 import numpy as np
 
-from .builders import random_circuit
+from .builders import random_circuit, sycamore_circuit
 from .connectivity import ConnectivityLayout
 from .tensor import CompositeTensor, LeafTensor, TensorData
 
@@ -31,6 +31,9 @@ FIXTURES = {
     # instance MFMA-bound (large intermediates) while fitting one 288 GB GPU
     "rqc36": dict(qubits=36, rounds=14, p1=0.5, p2=0.8, seed=52,
                   layout=ConnectivityLayout.EAGLE, trials=64, size_cap=6.0e9),
+    # 49-qubit Sycamore-style RQC single amplitude (config 5, complex64)
+    "syc49": dict(kind="sycamore", qubits=49, depth=12, seed=42, trials=48,
+                  size_cap=8.0e9, dtype="c64"),
 }
 
 
@@ -84,6 +87,9 @@ def load_network(path: str):
 
 def build_fixture(name: str) -> CompositeTensor:
     cfg = FIXTURES[name]
+    if cfg.get("kind") == "sycamore":
+        c = sycamore_circuit(cfg["qubits"], cfg["depth"], cfg["seed"])
+        return c.into_amplitude_network("0" * cfg["qubits"])[0]
     return random_circuit(cfg["qubits"], cfg["rounds"], cfg["p1"], cfg["p2"],
                           cfg["seed"], cfg["layout"])
 
