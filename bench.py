@@ -21,6 +21,10 @@ import os
 import sys
 import time
 
+# cap BLAS threads before numpy loads (cpu_baseline; see _blas_threads)
+os.environ.setdefault("OPENBLAS_NUM_THREADS", "64")
+os.environ.setdefault("OMP_NUM_THREADS", "64")
+
 ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, ROOT)
 
@@ -97,6 +101,16 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds, dtype="c128",
     }
 
 
+def _blas_threads():
+    try:
+        cores = len(os.sched_getaffinity(0))
+    except AttributeError:
+        cores = os.cpu_count() or 1
+    # OpenBLAS misbehaves (and slows down) beyond ~64 threads; the thread
+    # count actually used is reported in the JSON ("cores")
+    return min(64, cores)
+
+
 def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29,
                  dtype="c128"):
     """Oracle (numpy einsum -> BLAS zgemm) timed on the host cores over a
@@ -143,14 +157,10 @@ def cpu_baseline(infos, fixture="rqc36", budget_s=20.0, cap_elems=2 ** 29,
             break
     if total_time == 0.0:
         return None
-    try:
-        cores = len(os.sched_getaffinity(0))
-    except AttributeError:
-        cores = os.cpu_count()
     return {
         "value": total_flops / total_time / 1e9,
         "unit": "GFLOP/s",
-        "cores": cores,
+        "cores": _blas_threads(),
         "kind": "port",
         "sample": f"{used} largest {fixture} path steps ({shrunk} sampled "
                   f"at reduced M to fit {cap_elems} elems), random-valued "
@@ -187,9 +197,10 @@ def run_single(args):
     wall = time.perf_counter() - t0
 
     value = flops_per_contraction * args.steps / wall / 1e9
+    infos = eng.infos
+    eng.close()  # release the device arena before the host-BLAS baseline
     cb = None if args.no_cpu_baseline else cpu_baseline(
-        eng.infos, args.fixture, dtype=dtype)
-    eng.close()
+        infos, args.fixture, dtype=dtype)
     emit({
         "metric": f"pairwise-contraction GFLOP/s ({dtype})",
         "value": value,
