@@ -172,3 +172,63 @@ def test_eagle_connectivity_shape():
     # IBM Eagle: 127 qubits
     assert max(nodes) == 126
     assert len(edges) == len(set(tuple(sorted(e)) for e in edges))
+
+
+def test_random_circuit_with_set_observable_pinned_legs():
+    """random_circuit.rs:358-418: with p=1 and observable on qubit 2, the
+    leg layout is fully pinned (RNG-independent)."""
+    from tnc_amd.builders import random_circuit_with_set_observable
+
+    tn = random_circuit_with_set_observable(
+        4, 3, 1.0, 1.0, [2], 0, ConnectivityLayout.Line(4))
+    ref = [
+        [0, 1],
+        [3, 4, 2, 0], [2, 1, 5, 6], [8, 9, 4, 7], [6, 7, 10, 11],
+        [12, 3], [5, 13], [14, 8], [10, 15], [16, 9], [11, 17],
+        [19, 20, 18, 12], [18, 13, 21, 22], [23, 24, 20, 14],
+        [22, 15, 25, 26], [27, 28, 24, 16], [26, 17, 29, 30],
+        [31, 19], [21, 32], [33, 23], [25, 34], [35, 27], [29, 36],
+        [37, 28], [30, 38],
+        [31], [32], [33], [34], [35], [36], [37], [38],
+    ]
+    assert len(tn.tensors) == 33
+    for t, legs in zip(tn.tensors, ref):
+        assert t.legs == legs, (t.legs, legs)
+
+
+def test_random_circuit_with_observable_pinned_legs():
+    """random_circuit.rs:287-355: all probabilities 1 -> 40 tensors with the
+    reference's exact leg layout."""
+    from tnc_amd.builders import random_circuit_with_observable
+
+    tn = random_circuit_with_observable(
+        4, 3, 1.0, 1.0, 1.0, 1, ConnectivityLayout.Line(4))
+    assert len(tn.tensors) == 40
+    assert tn.tensors[4].legs == [8, 9, 0, 2]
+    assert tn.tensors[5].legs == [1, 3, 10, 11]
+    assert tn.tensors[-1].legs == [47]
+
+
+def test_observable_network_contracts():
+    """The sandwich network contracts to a scalar, path-independently (the
+    reference pairs each side with the SAME unconjugated random state,
+    random_circuit.rs:262-270, so the value is complex in general)."""
+    from tnc_amd.builders import random_circuit_with_set_observable
+
+    tn = random_circuit_with_set_observable(
+        4, 3, 0.7, 0.7, [1, 2], 3, ConnectivityLayout.Line(4))
+    out = oracle_contract(tn, RandomGreedy(4))
+    assert out.legs == []
+    out2 = oracle_contract(tn, Greedy())
+    np.testing.assert_allclose(out.data, out2.data, rtol=1e-12)
+
+
+def test_is_connected():
+    from tnc_amd import CompositeTensor, LeafTensor
+
+    bd = {0: 17, 1: 19, 2: 8, 3: 2}
+    tn = CompositeTensor([LeafTensor.new_from_map([0, 1], bd),
+                          LeafTensor.new_from_map([1, 2], bd)])
+    assert tn.is_connected()
+    tn.push_tensor(LeafTensor.new_from_map([3], bd))
+    assert not tn.is_connected()

@@ -24,6 +24,7 @@ from .paths import Greedy, Optimal, RandomGreedy, BasicContractionPathResult
 from .circuit import Circuit, Permutor
 from .connectivity import ConnectivityLayout, connectivity_edges
 from .builders import random_circuit, sycamore_circuit
+from .qasm import import_qasm
 from .partition import find_partitioning, partition_tensor_network
 from .repartition import (
     CommunicationScheme,

@@ -80,3 +80,110 @@ def sycamore_circuit(qubits, depth, rng) -> Circuit:
                     [qr.qubit(i - 1), qr.qubit(j - 1)],
                 )
     return circuit
+
+
+def random_circuit_with_observable(
+    qubits, rounds, single_qubit_probability, two_qubit_probability,
+    observable_probability, rng, connectivity
+) -> CompositeTensor:
+    """random_circuit.rs:88-113: random observable placement, then the
+    set-observable construction."""
+    rng = _rng(rng)
+    locations = [i for i in range(qubits) if rng.random() < observable_probability]
+    return random_circuit_with_set_observable(
+        qubits, rounds, single_qubit_probability, two_qubit_probability,
+        locations, rng, connectivity,
+    )
+
+
+def random_circuit_with_set_observable(
+    qubits, rounds, single_qubit_probability, two_qubit_probability,
+    observable_location, rng, connectivity
+) -> CompositeTensor:
+    """random_circuit.rs:120-275: a sandwich network <psi| U' O U |psi> built
+    directly — per qubit a (left, right) open-edge pair, observables first,
+    then mirrored fsim/single-qubit layers on qubits that affect the
+    observable, then a random shared initial state on each side.
+
+    Edge numbering reproduces the reference exactly (pinned by the
+    leg-layout tests at random_circuit.rs:287-418)."""
+    from .tensor import LeafTensor
+    from .tensorgeneration import random_sparse_tensor_data
+
+    rng = _rng(rng)
+    single_gates = [("sx", "sx"), ("sy", "sx"), ("sz", "sx")]  # (fwd, adj) per :132-145
+    observables = ["x", "y", "z"]
+
+    tn = CompositeTensor()
+    open_edges = {}
+    next_edge = 0
+
+    final_state = []
+    for i in range(qubits):
+        if i in observable_location:
+            open_edges[i] = (next_edge, next_edge + 1)
+            next_edge += 2
+            obs = observables[int(rng.integers(0, 3))]
+            t = LeafTensor.new_from_const([open_edges[i][0], open_edges[i][1]], 2)
+            t.set_tensor_data(TensorData.from_gate(obs))
+            final_state.append(t)
+        else:
+            open_edges[i] = (0, 0)
+    tn.push_tensors(final_state)
+
+    edges = [(u, v) for (u, v) in connectivity_edges(connectivity)
+             if u < qubits and v < qubits]
+    gates = []
+    for _ in range(1, rounds):
+        for i, j in edges:
+            if (rng.random() < two_qubit_probability
+                    and (open_edges[i][0] != open_edges[i][1]
+                         or open_edges[j][0] != open_edges[j][1])):
+                if open_edges[i][0] != open_edges[i][1]:
+                    li, ri = open_edges[i]
+                else:
+                    li = ri = next_edge
+                    next_edge += 1
+                if open_edges[j][0] != open_edges[j][1]:
+                    lj, rj = open_edges[j]
+                else:
+                    lj = rj = next_edge
+                    next_edge += 1
+                left = LeafTensor.new_from_const(
+                    [next_edge, next_edge + 1, li, lj], 2)
+                left.set_tensor_data(TensorData.from_gate("fsim", [0.3, 0.2]))
+                gates.append(left)
+                right = LeafTensor.new_from_const(
+                    [ri, rj, next_edge + 2, next_edge + 3], 2)
+                right.set_tensor_data(
+                    TensorData.from_gate("fsim", [0.3, 0.2], adjoint=True))
+                gates.append(right)
+                open_edges[i] = (next_edge, next_edge + 2)
+                open_edges[j] = (next_edge + 1, next_edge + 3)
+                next_edge += 4
+        for i in range(qubits):
+            li, ri = open_edges[i]
+            if rng.random() < single_qubit_probability and li != ri:
+                fwd, adj = single_gates[int(rng.integers(0, 3))]
+                left = LeafTensor.new_from_const([next_edge, li], 2)
+                left.set_tensor_data(TensorData.from_gate(fwd))
+                gates.append(left)
+                right = LeafTensor.new_from_const([ri, next_edge + 1], 2)
+                right.set_tensor_data(TensorData.from_gate(adj, adjoint=True))
+                gates.append(right)
+                open_edges[i] = (next_edge, next_edge + 1)
+                next_edge += 2
+    tn.push_tensors(gates)
+
+    initial = []
+    for i in range(qubits):
+        li, ri = open_edges[i]
+        if li != ri:
+            state = random_sparse_tensor_data([2], 1.0, rng)
+            left = LeafTensor.new_from_const([li], 2)
+            left.set_tensor_data(state)
+            right = LeafTensor.new_from_const([ri], 2)
+            right.set_tensor_data(state)
+            initial.extend([left, right])
+    tn.push_tensors(initial)
+    return tn

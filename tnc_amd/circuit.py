@@ -100,6 +100,11 @@ class Circuit:
     def num_qubits(self):
         return len(self.open_edges)
 
+    def qubit(self, index) -> _Qubit:
+        """Absolute qubit handle across registers (importer convenience)."""
+        assert 0 <= index < self.num_qubits()
+        return _Qubit(index)
+
     def _new_edge(self):
         e = self.next_edge
         self.next_edge += 1

@@ -212,6 +212,22 @@ class CompositeTensor:
             for t in self._tensors
         )
 
+    def is_connected(self) -> bool:
+        """All children pairwise-connected through shared legs
+        (tensor.rs:368-389; children must be leaves)."""
+        from .utils import UnionFind
+
+        n = len(self._tensors)
+        uf = UnionFind(n)
+        for i in range(n):
+            for j in range(i + 1, n):
+                t1, t2 = self._tensors[i], self._tensors[j]
+                assert not isinstance(t1, CompositeTensor)
+                assert not isinstance(t2, CompositeTensor)
+                if (t1 & t2).legs:
+                    uf.union(i, j)
+        return uf.count_sets() == 1
+
     def external_tensor(self) -> LeafTensor:
         """Open legs after full contraction (tensor.rs:392-403): fold of
         symmetric differences over children, in order."""
