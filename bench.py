@@ -66,8 +66,9 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds, dtype="c128",
         try:
             with open(pmc_path) as f:
                 pmc = json.load(f)
-            if pmc.get("fixture") == fixture:
-                traffic = pmc.get("dominant_kernel_traffic_bytes")
+            entry = pmc.get("fixtures", {}).get(fixture)
+            if entry:
+                traffic = entry.get("dominant_kernel_traffic_bytes")
         except Exception:
             traffic = None
     mfma_peak = F64_MFMA_PEAK if dtype == "c128" else F32_MFMA_PEAK
