@@ -64,3 +64,59 @@ def test_memory_limit_rejects():
     part = find_partitioning(tn, k)
     model = NaivePartitioningModel(tn, k, memory_limit=1.0)  # 1 byte
     assert model.evaluate(part, np.random.default_rng(0)) == float("inf")
+
+
+def test_leaf_model_sa():
+    from tnc_amd.repartition import LeafPartitioningModel, balance_partitions
+
+    tn = _net()
+    k = 3
+    initial = find_partitioning(tn, k)
+    model = LeafPartitioningModel(tn)
+    sol0 = model.initial_solution(initial)
+    rng = np.random.default_rng(1)
+    s0 = model.evaluate(sol0, rng)
+    best, score = balance_partitions(model, sol0, np.random.default_rng(1),
+                                     max_time_s=1.0, n_trials=2, n_steps=6)
+    assert score <= s0
+    ptn, path, _, _ = compute_solution(tn, best[0])
+    ref = contract_network(
+        network_to_otensors(tn),
+        __import__("tnc_amd").Greedy().find_path(tn).replace_path(),
+    )
+    out = contract_network(network_to_otensors(ptn), path)
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)
+
+
+def test_intermediate_model_sa():
+    """Config 4's IAD method (benchmark/src/main.rs:624-680)."""
+    from tnc_amd.repartition import (IntermediatePartitioningModel,
+                                     balance_partitions)
+
+    tn = _net()
+    k = 3
+    initial = find_partitioning(tn, k)
+    model = IntermediatePartitioningModel(tn)
+    sol0 = model.compute_initial_solution(initial)
+    rng = np.random.default_rng(2)
+    s0 = model.evaluate(sol0, rng)
+    best, score = balance_partitions(model, sol0, np.random.default_rng(2),
+                                     max_time_s=1.5, n_trials=2, n_steps=6)
+    assert score <= s0
+    # moved-subtree bookkeeping stays consistent: views match partitioning
+    part, views, paths = best
+    from tnc_amd.tensor import LeafTensor as LT
+
+    for p in range(k):
+        acc = LT([], [])
+        for t, pi in zip(tn.tensors, part):
+            if pi == p:
+                acc = acc ^ LT(t.legs, t.bond_dims)
+        assert sorted(acc.legs) == sorted(views[p].legs)
+    ptn, path, _, _ = compute_solution(tn, part)
+    ref = contract_network(
+        network_to_otensors(tn),
+        __import__("tnc_amd").Greedy().find_path(tn).replace_path(),
+    )
+    out = contract_network(network_to_otensors(ptn), path)
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)

@@ -30,7 +30,7 @@ from .cost import (
 )
 from .partition import partition_tensor_network
 from .paths import Greedy, RandomGreedy
-from .tensor import CompositeTensor
+from .tensor import CompositeTensor, LeafTensor
 
 
 class CommunicationScheme:
@@ -120,6 +120,149 @@ class NaivePartitioningModel:
 
     def evaluate(self, solution, rng):
         return _evaluate_partitioning(self.tensor, solution, self.scheme,
+                                      self.memory_limit, rng)
+
+
+class LeafPartitioningModel:
+    """Move a random tensor to the partition that minimizes
+    (shifted ^ partition).size() - partition.size()
+    (simulated_annealing.rs:351-404). Solution = (partitioning,
+    per-partition external views)."""
+
+    def __init__(self, tensor, scheme=CommunicationScheme.GREEDY,
+                 memory_limit=None):
+        self.tensor = tensor
+        self.scheme = scheme
+        self.memory_limit = memory_limit
+
+    def initial_solution(self, partitioning):
+        k = max(partitioning) + 1
+        views = []
+        for p in range(k):
+            acc = LeafTensor([], [])
+            for t, part in zip(self.tensor.tensors, partitioning):
+                if part == p:
+                    acc = acc ^ LeafTensor(t.legs, t.bond_dims)
+            views.append(acc)
+        return (list(partitioning), views)
+
+    def generate_trial_solution(self, solution, rng):
+        partitioning, views = list(solution[0]), list(solution[1])
+        idx = int(rng.integers(0, len(partitioning)))
+        t = self.tensor.tensors[idx]
+        shifted = LeafTensor(t.legs, t.bond_dims)
+        src = partitioning[idx]
+        best_p, best_gain = None, math.inf
+        for p, view in enumerate(views):
+            if p == src:
+                continue
+            gain = (shifted ^ view).size() - view.size()
+            if gain < best_gain:
+                best_p, best_gain = p, gain
+        partitioning[idx] = best_p
+        views[src] = views[src] ^ shifted
+        views[best_p] = views[best_p] ^ shifted
+        return (partitioning, views)
+
+    def evaluate(self, solution, rng):
+        return _evaluate_partitioning(self.tensor, solution[0], self.scheme,
+                                      self.memory_limit, rng)
+
+
+class IntermediatePartitioningModel:
+    """Move a random contraction subtree to the partition that minimizes the
+    memory-gain heuristic (simulated_annealing.rs:407-570; config 4's IAD
+    method). Solution = (partitioning, partition external views,
+    per-partition replace-left paths)."""
+
+    def __init__(self, tensor, scheme=CommunicationScheme.GREEDY,
+                 memory_limit=None):
+        self.tensor = tensor
+        self.scheme = scheme
+        self.memory_limit = memory_limit
+
+    def compute_initial_solution(self, partitioning):
+        """simulated_annealing.rs:416-452. The partitioning is normalized to
+        first-appearance ids so that partition VALUES equal the positions of
+        partition_tensor_network's children (the views/paths index space)."""
+        from .partition import partition_tensor_network
+        from .paths import Greedy
+
+        order = []
+        for p in partitioning:
+            if p not in order:
+                order.append(p)
+        remap = {p: i for i, p in enumerate(order)}
+        partitioning = [remap[p] for p in partitioning]
+        ptn = partition_tensor_network(self.tensor, partitioning)
+        views = [
+            t.external_tensor() if isinstance(t, CompositeTensor) else
+            LeafTensor(t.legs, t.bond_dims)
+            for t in ptn.tensors
+        ]
+        paths = []
+        for t in ptn.tensors:
+            if isinstance(t, CompositeTensor) and len(t.tensors) > 1:
+                paths.append(Greedy().find_path(t).replace_path().toplevel)
+            else:
+                paths.append([])
+        return (list(partitioning), views, paths)
+
+    def generate_trial_solution(self, solution, rng):
+        from .paths import Greedy
+
+        partitioning = list(solution[0])
+        views = list(solution[1])
+        paths = [list(p) for p in solution[2]]
+        viable = [p for p, path in enumerate(paths) if len(path) >= 3]
+        if not viable:
+            return (partitioning, views, paths)
+        src = viable[int(rng.integers(0, len(viable)))]
+        pair_index = int(rng.integers(0, len(paths[src]) - 1))
+        i, j = paths[src][pair_index]
+        # gather the subtree feeding this contraction
+        # (simulated_annealing.rs:485-496)
+        leaves = {i, j}
+        for a, b in reversed(paths[src][:pair_index]):
+            if a in leaves:
+                leaves.add(b)
+        shifted = LeafTensor([], [])
+        shifted_idx = []
+        local = 0
+        for gidx, part in enumerate(partitioning):
+            if part != src:
+                continue
+            if local in leaves:
+                t = self.tensor.tensors[gidx]
+                shifted = shifted ^ LeafTensor(t.legs, t.bond_dims)
+                shifted_idx.append(gidx)
+            local += 1
+        best_p, best_gain = None, math.inf
+        for p, view in enumerate(views):
+            if p == src:
+                continue
+            gain = (shifted ^ view).size() - view.size()
+            if gain < best_gain:
+                best_p, best_gain = p, gain
+        for gidx in shifted_idx:
+            partitioning[gidx] = best_p
+        views[src] = views[src] ^ shifted
+        views[best_p] = views[best_p] ^ shifted
+        # recompute local paths of both partitions
+        # (simulated_annealing.rs:538-560)
+        for part in (src, best_p):
+            sub = CompositeTensor([
+                t for t, pi in zip(self.tensor.tensors, partitioning)
+                if pi == part
+            ])
+            if len(sub.tensors) > 1:
+                paths[part] = Greedy().find_path(sub).replace_path().toplevel
+            else:
+                paths[part] = []
+        return (partitioning, views, paths)
+
+    def evaluate(self, solution, rng):
+        return _evaluate_partitioning(self.tensor, solution[0], self.scheme,
                                       self.memory_limit, rng)
 
 
