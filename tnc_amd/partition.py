@@ -44,12 +44,13 @@ def _build_graph(tn: CompositeTensor):
 
 
 def find_partitioning(tn: CompositeTensor, k: int, strategy=PartitioningStrategy.MIN_CUT,
-                      minimize=True, seed=0) -> list:
+                      minimize=True, seed=0, imbalance=0.03) -> list:
     """Assign each tensor a partition id in [0, k) (partitioning.rs:31-90).
 
     BFS region growing to balanced sizes, then greedy boundary refinement
     minimizing the cut weight; imbalance tolerance 3% like the reference
-    (partitioning.rs:47).
+    (partitioning.rs:47), overridable (communication_partitioning passes its
+    own, partitioning.rs:100-106).
     """
     n = len(tn.tensors)
     if k <= 1:
@@ -58,7 +59,7 @@ def find_partitioning(tn: CompositeTensor, k: int, strategy=PartitioningStrategy
     rng = np.random.Generator(np.random.PCG64(seed))
 
     target = n / k
-    cap = int(math.ceil(target * 1.03))
+    cap = int(math.ceil(target * (1.0 + imbalance)))
     part = [-1] * n
     # seed each region with a far-apart start (greedy: highest-degree unused)
     order = sorted(range(n), key=lambda t: -sum(adj[t].values()))

@@ -36,27 +36,63 @@ from .tensor import CompositeTensor, LeafTensor
 class CommunicationScheme:
     GREEDY = "greedy"
     RANDOM_GREEDY = "random_greedy"
+    BIPARTITION = "bipartition"
+    BIPARTITION_SWEEP = "bipartition_sweep"
+
+
+def _tensor_bipartition_recursive(children, imbalance):
+    """communication_schemes.rs:147-205: recursive 2-way min-cut of the
+    boundary tensors; the larger side keeps the result slot. Returns
+    (root id, folded view, path)."""
+    from .partition import find_partitioning
+
+    if len(children) == 1:
+        i, t = children[0]
+        return i, t, []
+    if len(children) == 2:
+        (i0, t0), (i1, t1) = children
+        a, b = (i1, i0) if t1.size() > t0.size() else (i0, i1)
+        return a, t0 ^ t1, [(a, b)]
+    tn = CompositeTensor([t for _, t in children])
+    part = find_partitioning(tn, 2, imbalance=imbalance)
+    left = [c for c, p in zip(children, part) if p == part[0]]
+    right = [c for c, p in zip(children, part) if p != part[0]]
+    if not right:  # degenerate cut: split arbitrarily to guarantee progress
+        left, right = children[: len(children) // 2], children[len(children) // 2:]
+    i1, t1, p1 = _tensor_bipartition_recursive(left, imbalance)
+    i2, t2, p2 = _tensor_bipartition_recursive(right, imbalance)
+    out = t1 ^ t2
+    a, b = (i2, i1) if t2.size() > t1.size() else (i1, i2)
+    return a, out, p1 + p2 + [(a, b)]
 
 
 def communication_path(children, latency_map, scheme, rng=None, trials=8):
-    """Fan-in path over partition externals (communication_schemes.rs:49-80).
-
-    greedy: plain greedy over the externals (:75-80); random_greedy:
-    randomized greedy trials, best by latency-aware critical-path cost."""
+    """Fan-in path over partition externals (communication_schemes.rs:49-80):
+    greedy (:75-80), random_greedy = RandomGreedy(100) (:215-221),
+    bipartition (imbalance 0.03, :82-88), bipartition_sweep (20 random
+    imbalances, best latency-aware critical-path cost, :90-123)."""
     tn = CompositeTensor(list(children))
     if scheme == CommunicationScheme.GREEDY:
         return Greedy().find_path(tn).replace_path().toplevel
-    best = None
-    best_cost = math.inf
-    latencies = [latency_map.get(i, 0.0) for i in range(len(children))]
-    seed = int(rng.integers(0, 2**31)) if rng is not None else 42
-    for t in range(max(1, trials)):
-        path = RandomGreedy(2, seed=seed + t).find_path(tn).replace_path().toplevel
-        cost, _ = communication_path_cost(children, path, True, True, latencies)
-        if cost < best_cost:
-            best_cost = cost
-            best = path
-    return best
+    if scheme == CommunicationScheme.RANDOM_GREEDY:
+        return RandomGreedy(100).find_path(tn).replace_path().toplevel
+    indexed = [(i, LeafTensor(t.legs, t.bond_dims))
+               for i, t in enumerate(children)]
+    if scheme == CommunicationScheme.BIPARTITION:
+        return _tensor_bipartition_recursive(indexed, 0.03)[2]
+    if scheme == CommunicationScheme.BIPARTITION_SWEEP:
+        assert rng is not None, "BipartitionSweep requires a rng"
+        latencies = [latency_map.get(i, 0.0) for i in range(len(children))]
+        best, best_cost = None, math.inf
+        for _ in range(20):
+            imb = float(rng.uniform(0.01, 0.5))
+            path = _tensor_bipartition_recursive(indexed, imb)[2]
+            cost, _ = communication_path_cost(children, path, True, True,
+                                              latencies)
+            if cost < best_cost:
+                best, best_cost = path, cost
+        return best
+    raise ValueError(f"unknown communication scheme {scheme!r}")
 
 
 def compute_solution(tensor, partitioning, scheme=CommunicationScheme.GREEDY,
