@@ -122,6 +122,173 @@ __global__ __launch_bounds__(WAVES * 64) void zg(
     }
 }
 
+
+// glds variant: LDS-DMA staging (no staging registers / ds_writes),
+// interleaved c128 LDS image with XOR swizzle, operands read as
+// ds_read_b128 (re+im in one instruction). 128x64 tile, 8 waves, KT=16.
+__global__ __launch_bounds__(512) void zg_glds(
+    const double2* __restrict__ A, const double2* __restrict__ B,
+    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles) {
+  constexpr int TM = 128, TN = 64, KT = 16;
+  __shared__ double2 As[TM * KT];  // [r][c ^ (r & 15)]
+  __shared__ double2 Bs[KT * TN];  // [k][j ^ ((k & 3) << 4)]
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const u64 tile = blockIdx.x;
+  const u64 brow = (tile / col_tiles) * TM, bcol = (tile % col_tiles) * TN;
+  v4d cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) { cr[f] = v4d{0,0,0,0}; ci[f] = v4d{0,0,0,0}; }
+  const int fi = lane % 16;
+  const int fk = lane / 16;
+  // glds lane mapping: each wave-wide glds writes 64 consecutive 16B LDS
+  // slots starting at a wave-uniform base; lane l supplies the element that
+  // belongs at base + l. A tile: 128*16 = 2048 slots = 4 glds per wave
+  // (8 waves): wave w, piece p covers rows [ (w*4+p)*... ]. Linear LDS index
+  // i = r*KT + c_sw; we choose src so that image[i] = A[r][c] with
+  // c = c_sw ^ (r & 15).
+  for (u64 k0 = 0; k0 < K; k0 += KT) {
+    for (int piece = 0; piece < 4; ++piece) {
+      // LDS dst = wave-uniform base + lane*16 (implicit); choose the SOURCE
+      // per lane so the swizzled image lands linearly.
+      int base = (wave * 4 + piece) * 64;
+      int i = base + lane;
+      int r = i / KT, c_sw = i % KT;
+      int c = c_sw ^ (r & 15);
+      const double2* src = &A[(brow + r) * K + k0 + c];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&As[base], 16, 0, 0);
+    }
+    for (int piece = 0; piece < 2; ++piece) {
+      int base = piece * 512 + wave * 64;
+      int j = base + lane;
+      int k = j / TN, col_sw = j % TN;
+      int col = col_sw ^ ((k & 3) << 4);
+      const double2* src = &B[(k0 + k) * N + bcol + col];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&Bs[base], 16, 0, 0);
+    }
+    __syncthreads();  // carries vmcnt(0): drains the DMAs
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + fk;
+      double2 a = As[arow * KT + (ak ^ (arow & 15))];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        double2 b = Bs[ak * TN + (bcolf ^ ((ak & 3) << 4))];
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow0 = wave * 16 + (lane / 16);
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f)
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + 4 * r;
+      u64 col = bcol + f * 16 + ccol;
+      if (row < M && col < N) C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+    }
+}
+
+static void launch_glds(const double2* A, const double2* B, double2* C, u64 M,
+                        u64 N, u64 K) {
+  u64 rt = (M + 127) / 128, ct = (N + 63) / 64;
+  hipLaunchKernelGGL(zg_glds, dim3((unsigned)(rt * ct)), dim3(512), 0, 0, A,
+                     B, C, M, N, K, ct);
+}
+
+
+// glds double-buffered: 2 LDS buffers, raw barrier, counted vmcnt — the
+// next K-tile's DMAs overlap the current tile's MFMAs. Wave w's A slots are
+// self-written (rows 16w..16w+16); B slots are cross-wave (barrier covers).
+__global__ __launch_bounds__(512) void zg_glds_db(
+    const double2* __restrict__ A, const double2* __restrict__ B,
+    double2* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles) {
+  constexpr int TM = 128, TN = 64, KT = 16;
+  constexpr int ASLOTS = TM * KT, BSLOTS = KT * TN;
+  __shared__ double2 lds[2 * (ASLOTS + BSLOTS)];
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const u64 tile = blockIdx.x;
+  const u64 brow = (tile / col_tiles) * TM, bcol = (tile % col_tiles) * TN;
+  v4d cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) { cr[f] = v4d{0,0,0,0}; ci[f] = v4d{0,0,0,0}; }
+  const int fi = lane % 16;
+  const int fk = lane / 16;
+
+  auto stage = [&](int buf, u64 k0) {
+    double2* As = lds + buf * (ASLOTS + BSLOTS);
+    double2* Bs = As + ASLOTS;
+    for (int piece = 0; piece < 4; ++piece) {
+      int base = (wave * 4 + piece) * 64;
+      int i = base + lane;
+      int r = i / KT, c_sw = i % KT;
+      int c = c_sw ^ (r & 15);
+      const double2* src = &A[(brow + r) * K + k0 + c];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&As[base], 16, 0, 0);
+    }
+    for (int piece = 0; piece < 2; ++piece) {
+      int base = piece * 512 + wave * 64;
+      int j = base + lane;
+      int k = j / TN, col_sw = j % TN;
+      int col = col_sw ^ ((k & 3) << 4);
+      const double2* src = &B[(k0 + k) * N + bcol + col];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&Bs[base], 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  int buf = 0;
+  for (u64 k0 = 0; k0 < K; k0 += KT) {
+    if (k0 + KT < K) stage(buf ^ 1, k0 + KT);
+    // wait for the CURRENT buffer's own 6 DMAs (the 6 just issued for the
+    // next buffer stay in flight), then sync all waves (B is cross-wave)
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const double2* As = lds + buf * (ASLOTS + BSLOTS);
+    const double2* Bs = As + ASLOTS;
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + fk;
+      double2 a = As[arow * KT + (ak ^ (arow & 15))];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        double2 b = Bs[ak * TN + (bcolf ^ ((ak & 3) << 4))];
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+  const int crow0 = wave * 16 + (lane / 16);
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f)
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + 4 * r;
+      u64 col = bcol + f * 16 + ccol;
+      if (row < M && col < N) C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+    }
+}
+
+static void launch_glds_db(const double2* A, const double2* B, double2* C,
+                           u64 M, u64 N, u64 K) {
+  u64 rt = (M + 127) / 128, ct = (N + 63) / 64;
+  hipLaunchKernelGGL(zg_glds_db, dim3((unsigned)(rt * ct)), dim3(512), 0, 0,
+                     A, B, C, M, N, K, ct);
+}
+
 struct Variant {
   const char* name;
   void (*launch)(const double2*, const double2*, double2*, u64, u64, u64);
@@ -154,13 +321,19 @@ int main(int argc, char** argv) {
       static __global__ void f(double2* p, u64 n, int seed) {
         for (u64 i = blockIdx.x * (u64)blockDim.x + threadIdx.x; i < n;
              i += gridDim.x * (u64)blockDim.x) {
+          // FULL-entropy mantissas: low-entropy values flatter DVFS-bound
+          // kernels by ~20% (MI355X_MICROARCH.md "DVFS give-back")
           unsigned x = (unsigned)(i * 2654435761u) ^ (seed * 40503u);
           x ^= x >> 13;
           x *= 0x5bd1e995u;
           x ^= x >> 15;
           unsigned y = x * 1664525u + 1013904223u;
-          p[i] = make_double2((double)(x & 0xffff) / 65536.0 - 0.5,
-                              (double)(y & 0xffff) / 65536.0 - 0.5);
+          unsigned z = y * 22695477u + 1u;
+          p[i] = make_double2(
+              ((double)x + (double)y * 2.3283064365386963e-10) /
+                      4294967296.0 - 0.5,
+              ((double)y + (double)z * 2.3283064365386963e-10) /
+                      4294967296.0 - 0.5);
         }
       }
     };
@@ -172,10 +345,8 @@ int main(int argc, char** argv) {
 
   Variant variants[] = {
       {"w8 k16 (base)", launch_zg<8, 16, 0>},
-      {"w8 k16 gc8", launch_zg<8, 16, 0, 4, 8>},
-      {"w8 k16 gc16", launch_zg<8, 16, 0, 4, 16>},
-      {"w8 k16 gc32", launch_zg<8, 16, 0, 4, 32>},
-      {"w8 k16 f8 gc8", launch_zg<8, 16, 0, 8, 8>},
+      {"w8 k16 glds", launch_glds},
+      {"w8 k16 glds-db", launch_glds_db},
   };
   double flops = 8.0 * M * N * K;
   hipEvent_t e0, e1;

@@ -27,6 +27,7 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <type_traits>
 #include <vector>
 
 #include "../../include/tnc_hip.h"
@@ -229,17 +230,19 @@ __global__ void k_permute_ct(const CT* __restrict__ src,
 template <typename CT>
 __global__ __launch_bounds__(256) void k_zgemm_v1(
     const CT* __restrict__ A, const CT* __restrict__ B,
-    CT* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
-    u64 kchunk) {
+    CT* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk) {
   __shared__ CT As[GT][GK + 1];
   __shared__ CT Bs[GK][GT + 1];
   const int tx = threadIdx.x % 16, ty = threadIdx.x / 16;
-  const u64 tile = blockIdx.x % tiles;
-  const u64 slice = blockIdx.x / tiles;
-  const u64 brow = (tile / col_tiles) * GT, bcol = (tile % col_tiles) * GT;
-  const u64 kbeg = slice * kchunk;
+  // 32-bit tile decode (64-bit div emulation costs ~20 VGPRs)
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * GT;
+  const u64 bcol = (u64)(tile % col_tiles) * GT;
+  const u64 kbeg = (u64)slice * kchunk;
   const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
-  C += slice * M * N;  // slice 0 == C itself when kchunk == K
+  C += (u64)slice * M * N;  // slice 0 == C itself when kchunk == K
   CT acc[4][4];
   for (int i = 0; i < 4; ++i)
     for (int j = 0; j < 4; ++j) acc[i][j] = CT{0, 0};
@@ -324,8 +327,8 @@ struct MfmaCore<float> {
 template <typename CT>
 __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
     const CT* __restrict__ A, const CT* __restrict__ B,
-    CT* __restrict__ C, u64 M, u64 N, u64 K, u64 col_tiles, u64 tiles,
-    u64 kchunk) {
+    CT* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk) {
   using RT = decltype(CT{}.x);
   using Core = MfmaCore<RT>;
   using acc_t = typename Core::acc_t;
@@ -336,12 +339,13 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
-  const u64 tile = blockIdx.x % tiles;
-  const u64 slice = blockIdx.x / tiles;
-  const u64 brow = (tile / col_tiles) * MF_T, bcol = (tile % col_tiles) * MF_TN;
-  const u64 kbeg = slice * kchunk;
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * MF_T;
+  const u64 bcol = (u64)(tile % col_tiles) * MF_TN;
+  const u64 kbeg = (u64)slice * kchunk;
   const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
-  C += slice * M * N;
+  C += (u64)slice * M * N;
   const bool interior = (brow + MF_T <= M) && (bcol + MF_TN <= N);
 
   acc_t cr[4], ci[4];
@@ -398,6 +402,185 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_mfma(
         C[row * N + col] = CT{cr[f][r], ci[f][r]};
     }
   }
+}
+
+
+// c128 glds kernel: same 128x64 tile, but LDS-DMA staging into an XOR-
+// swizzled interleaved image read back as single ds_read_b128 per operand
+// element (re+im together). +20% over the planar-staged kernel on the
+// dominant rqc36 shape (70.2-70.7 TF/s = 90% of the 78.6 TF/s f64 spec
+// peak; scripts/zgemm_tune.hip — double-buffering loses to occupancy).
+// Edge tiles and K-tails use a bounds-guarded staging loop writing the
+// identical image (glds cannot mask out-of-range lanes).
+__global__ __launch_bounds__(MF_THREADS) void k_zgemm_c128_glds(
+    const double2* __restrict__ A, const double2* __restrict__ B,
+    double2* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk) {
+  constexpr int TM = MF_T, TN = MF_TN, KT = MF_K;
+  constexpr int ASLOTS = TM * KT;
+  __shared__ double2 As[ASLOTS];  // [r][c ^ (r & 15)]
+  __shared__ double2 Bs[KT * TN]; // [k][j ^ ((k & 3) << 4)]
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * TM;
+  const u64 bcol = (u64)(tile % col_tiles) * TN;
+  const u64 kbeg = (u64)slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += (u64)slice * M * N;
+  const bool interior = (brow + TM <= M) && (bcol + TN <= N);
+
+  v4d cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) {
+    cr[f] = v4d{0, 0, 0, 0};
+    ci[f] = v4d{0, 0, 0, 0};
+  }
+  const int fi = lane % 16;
+
+  for (u64 k0 = kbeg; k0 < kend; k0 += KT) {
+    if (interior && k0 + KT <= kend) {
+      // glds: LDS dst = wave-uniform base + lane*16; per-lane SOURCE
+      // pre-swizzled so the image lands linearly.
+      for (int piece = 0; piece < 4; ++piece) {
+        int base = (wave * 4 + piece) * 64;
+        int i = base + lane;
+        int r = i / KT, c_sw = i % KT;
+        int c = c_sw ^ (r & 15);
+        const double2* src = &A[(brow + r) * K + k0 + c];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)&As[base], 16, 0, 0);
+      }
+      for (int piece = 0; piece < 2; ++piece) {
+        int base = piece * 512 + wave * 64;
+        int j = base + lane;
+        int k = j / TN, col_sw = j % TN;
+        int col = col_sw ^ ((k & 3) << 4);
+        const double2* src = &B[(k0 + k) * N + bcol + col];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)src,
+            (__attribute__((address_space(3))) void*)&Bs[base], 16, 0, 0);
+      }
+    } else {
+      for (int i = threadIdx.x; i < TM * KT; i += MF_THREADS) {
+        int r = i / KT, c = i % KT;
+        double2 v = (brow + r < M && k0 + c < kend)
+                        ? A[(brow + r) * K + k0 + c]
+                        : make_double2(0.0, 0.0);
+        As[r * KT + (c ^ (r & 15))] = v;
+      }
+      for (int i = threadIdx.x; i < KT * TN; i += MF_THREADS) {
+        int k = i / TN, col = i % TN;
+        double2 v = (k0 + k < kend && bcol + col < N)
+                        ? B[(k0 + k) * N + bcol + col]
+                        : make_double2(0.0, 0.0);
+        Bs[k * TN + (col ^ ((k & 3) << 4))] = v;
+      }
+    }
+    __syncthreads();  // carries vmcnt(0): drains the LDS-DMAs
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + (lane / 16);
+      double2 a = As[arow * KT + (ak ^ (arow & 15))];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        double2 b = Bs[ak * TN + (bcolf ^ ((ak & 3) << 4))];
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int crow0 = wave * 16 + (lane / 16);
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f) {
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + 4 * r;
+      u64 col = bcol + f * 16 + ccol;
+      if (interior || (row < M && col < N))
+        C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+    }
+  }
+}
+
+
+// Pure-glds variant: no guarded branch in the loop (the in-loop fallback
+// costs ~20% via register pressure). Launched only when M % 128 == 0,
+// N % 64 == 0 and every K-slice is a whole number of 16-tiles.
+__global__ __launch_bounds__(MF_THREADS) void k_zgemm_c128_glds_pure(
+    const double2* __restrict__ A, const double2* __restrict__ B,
+    double2* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk) {
+  constexpr int TM = MF_T, TN = MF_TN, KT = MF_K;
+  constexpr int ASLOTS = TM * KT;
+  __shared__ double2 As[ASLOTS];
+  __shared__ double2 Bs[KT * TN];
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  // 32-bit tile decode: 64-bit div emulation costs ~22 VGPRs and a whole
+  // wave of occupancy (148 -> 126 VGPRs measured)
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * TM;
+  const u64 bcol = (u64)(tile % col_tiles) * TN;
+  const u64 kbeg = (u64)slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += (u64)slice * M * N;
+  v4d cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) {
+    cr[f] = v4d{0, 0, 0, 0};
+    ci[f] = v4d{0, 0, 0, 0};
+  }
+  const int fi = lane % 16;
+  for (u64 k0 = kbeg; k0 < kend; k0 += KT) {
+    for (int piece = 0; piece < 4; ++piece) {
+      int base = (wave * 4 + piece) * 64;
+      int i = base + lane;
+      int r = i / KT, c_sw = i % KT;
+      int c = c_sw ^ (r & 15);
+      const double2* src = &A[(brow + r) * K + k0 + c];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&As[base], 16, 0, 0);
+    }
+    for (int piece = 0; piece < 2; ++piece) {
+      int base = piece * 512 + wave * 64;
+      int j = base + lane;
+      int k = j / TN, col_sw = j % TN;
+      int col = col_sw ^ ((k & 3) << 4);
+      const double2* src = &B[(k0 + k) * N + bcol + col];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&Bs[base], 16, 0, 0);
+    }
+    __syncthreads();
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + (lane / 16);
+      double2 a = As[arow * KT + (ak ^ (arow & 15))];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        double2 b = Bs[ak * TN + (bcolf ^ ((ak & 3) << 4))];
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow0 = wave * 16 + (lane / 16);
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f)
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + 4 * r;
+      u64 col = bcol + f * 16 + ccol;
+      C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+    }
 }
 
 // split-K reduce: C[p] = sum over slices of ws[s][p]
@@ -848,12 +1031,28 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   dim3 grid((unsigned)(tiles * splitk));
   if (stats && stats->gemm_ev0)
     HIP_CHECK(hipEventRecord(stats->gemm_ev0, stream));
-  if (mfma)
-    k_zgemm_mfma<<<grid, MF_THREADS, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
-                                                  col_tiles, tiles, kchunk);
-  else
-    k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K, col_tiles,
-                                         tiles, kchunk);
+  if (mfma) {
+    if constexpr (std::is_same_v<CT, double2>) {
+      bool pure = (M % MF_T == 0) && (N % MF_TN == 0) && (K % MF_K == 0) &&
+                  (kchunk % MF_K == 0);
+      if (pure)
+        k_zgemm_c128_glds_pure<<<grid, MF_THREADS, 0, stream>>>(
+            Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
+            kchunk);
+      else
+        k_zgemm_c128_glds<<<grid, MF_THREADS, 0, stream>>>(
+            Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
+            kchunk);
+    } else {
+      k_zgemm_mfma<<<grid, MF_THREADS, 0, stream>>>(
+          Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
+          kchunk);
+    }
+  } else {
+    k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
+                                         (unsigned)col_tiles, (unsigned)tiles,
+                                         kchunk);
+  }
   if (splitk > 1)
     k_splitk_reduce<<<grid_for(nout), 256, 0, stream>>>(splitbuf, Cg, nout,
                                                         (int)splitk);
