@@ -583,6 +583,81 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_c128_glds_pure(
     }
 }
 
+
+// c64 pure-glds kernel: one 16-byte LDS-DMA moves TWO float2 elements, so
+// the swizzle works at pair granularity (8 pairs per 16-deep K row).
+__global__ __launch_bounds__(MF_THREADS) void k_zgemm_c64_glds_pure(
+    const float2* __restrict__ A, const float2* __restrict__ B,
+    float2* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk) {
+  constexpr int TM = MF_T, TN = MF_TN, KT = MF_K;
+  __shared__ float2 As[TM * KT];  // [r][2*(cp ^ (r & 7)) + e]
+  __shared__ float2 Bs[KT * TN];  // [k][2*(jp ^ ((k & 3) << 2)) + e]
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * TM;
+  const u64 bcol = (u64)(tile % col_tiles) * TN;
+  const u64 kbeg = (u64)slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += (u64)slice * M * N;
+  v4f cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) {
+    cr[f] = v4f{0, 0, 0, 0};
+    ci[f] = v4f{0, 0, 0, 0};
+  }
+  const int fi = lane % 16;
+  for (u64 k0 = kbeg; k0 < kend; k0 += KT) {
+    // A: 128x16 c64 = 1024 pair-slots; 8 waves x 2 pieces x 64 lanes
+    for (int piece = 0; piece < 2; ++piece) {
+      int base = (wave * 2 + piece) * 64;  // pair-slot base
+      int i = base + lane;
+      int r = i / (KT / 2), cp_sw = i % (KT / 2);
+      int cp = cp_sw ^ (r & 7);
+      const float2* src = &A[(brow + r) * K + k0 + 2 * cp];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&As[2 * base], 16, 0, 0);
+    }
+    // B: 16x64 c64 = 512 pair-slots; 8 waves x 1 piece
+    {
+      int base = wave * 64;
+      int j = base + lane;
+      int k = j / (TN / 2), jp_sw = j % (TN / 2);
+      int jp = jp_sw ^ ((k & 3) << 2);
+      const float2* src = &B[(k0 + k) * N + bcol + 2 * jp];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&Bs[2 * base], 16, 0, 0);
+    }
+    __syncthreads();
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + (lane / 16);
+      float2 a = As[arow * KT + 2 * ((ak / 2) ^ (arow & 7)) + (ak & 1)];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        float2 b = Bs[ak * TN + 2 * ((bcolf / 2) ^ ((ak & 3) << 2)) +
+                      (bcolf & 1)];
+        cr[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f32_16x16x4f32(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow0 = wave * 16 + (lane / 16) * 4;  // f32 D map: (l/16)*4+r
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f)
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + r;
+      u64 col = bcol + f * 16 + ccol;
+      C[row * N + col] = make_float2(cr[f][r], ci[f][r]);
+    }
+}
+
 // split-K reduce: C[p] = sum over slices of ws[s][p]
 template <typename CT>
 __global__ void k_splitk_reduce(const CT* __restrict__ ws,
@@ -1044,9 +1119,16 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
             Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
             kchunk);
     } else {
-      k_zgemm_mfma<<<grid, MF_THREADS, 0, stream>>>(
-          Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
-          kchunk);
+      bool pure = (M % MF_T == 0) && (N % MF_TN == 0) && (K % MF_K == 0) &&
+                  (kchunk % MF_K == 0);
+      if (pure)
+        k_zgemm_c64_glds_pure<<<grid, MF_THREADS, 0, stream>>>(
+            (const float2*)Ag, (const float2*)Bg, (float2*)gemm_out, M, N, K,
+            (unsigned)col_tiles, (unsigned)tiles, kchunk);
+      else
+        k_zgemm_mfma<<<grid, MF_THREADS, 0, stream>>>(
+            Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
+            kchunk);
     }
   } else {
     k_zgemm_v1<<<grid, 256, 0, stream>>>(Ag, Bg, gemm_out, M, N, K,
