@@ -234,7 +234,7 @@ def run_distributed(args):
 
     from tnc_amd import hiplib
     from tnc_amd.contraction_path import ContractionPath
-    from tnc_amd.dist import make_plan, run_fanin
+    from tnc_amd.dist import make_plan, make_tree_plan, run_fanin
     from tnc_amd.executor import ContractionEngine
     from tnc_amd.fixtures import load_fixture
     from tnc_amd.tensor import CompositeTensor
@@ -246,8 +246,16 @@ def run_distributed(args):
     dist_t.init_process_group("nccl")
     device = torch.device(f"cuda:{local_rank}")
 
-    tn, _, meta = load_fixture(args.fixture)
-    plan = make_plan(tn, world, trials=args.trials, size_cap=4.0e9)
+    tn, frozen_path, meta = load_fixture(args.fixture)
+    if frozen_path:
+        # cut the frozen contraction tree into `world` subtrees: total flops
+        # preserved, every exchanged tensor is a path intermediate (bounded
+        # memory); deterministic on every rank. A fresh k-way min-cut on
+        # amplitude networks can produce astronomically large boundary
+        # tensors (see tnc_amd/dist.py::make_tree_plan).
+        plan = make_tree_plan(tn, frozen_path, world)
+    else:
+        plan = make_plan(tn, world, trials=args.trials, size_cap=4.0e9)
     my_part = None
     for p, r in plan.part_rank.items():
         if r == rank:

@@ -129,3 +129,101 @@ def test_make_plan_rqc36_eight_way():
     plan2 = make_plan(tn, 8, trials=4, size_cap=4.0e9)
     assert plan2.part_rank == plan.part_rank
     assert plan2.path.toplevel == plan.path.toplevel
+
+
+def test_tree_plan_fanin_two_ranks_gloo():
+    """Tree-split plan executed across 2 gloo processes == direct result."""
+    _run_tree(2)
+
+
+def _tree_worker(rank, world, result_queue):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29519"
+    dist_t.init_process_group("gloo", rank=rank, world_size=world)
+
+    import numpy as np
+
+    from oracle import OTensor, contract_network, contract_tensors
+    from oracle.adapters import network_to_otensors
+    from tnc_amd.contraction_path import ContractionPath
+    from tnc_amd.dist import make_tree_plan, run_fanin
+    from tnc_amd.fixtures import load_fixture
+    from tnc_amd.tensor import CompositeTensor
+
+    tn, rp, _ = load_fixture("rqc24")
+    plan = make_tree_plan(tn, rp, world)
+
+    my_part = None
+    for p, r in plan.part_rank.items():
+        if r == rank:
+            my_part = p
+    local = None
+    if my_part is not None:
+        sub = plan.partitioned.tensors[my_part]
+        inner = plan.path.nested.get(my_part)
+        otensors = network_to_otensors(sub)
+        if inner is not None and inner.toplevel:
+            local = contract_network(otensors, inner)
+        else:
+            assert len(otensors) == 1
+            local = otensors[0]
+
+    def send(handle, legs, dims, peer):
+        arr = np.ascontiguousarray(handle.data, dtype=np.complex128)
+        t = torch.from_numpy(arr.view(np.float64).reshape(-1))
+        dist_t.send(t, dst=peer)
+
+    def recv(legs, dims, peer):
+        n = int(np.prod(dims)) if dims else 1
+        t = torch.empty(n * 2, dtype=torch.float64)
+        dist_t.recv(t, src=peer)
+        data = t.numpy().view(np.complex128).reshape([int(d) for d in dims])
+        return OTensor(list(legs), data)
+
+    def contract_pair(a, a_legs, a_dims, b, b_legs, b_dims):
+        assert list(a.legs) == list(a_legs)
+        return contract_tensors(a, b)
+
+    final = run_fanin(plan, rank, local, send, recv, contract_pair)
+    if rank == 0:
+        ref = contract_network(network_to_otensors(tn),
+                               ContractionPath.simple(rp))
+        ok = np.allclose(final.data, ref.data, rtol=1e-12)
+        result_queue.put("ok" if ok else "mismatch")
+    dist_t.destroy_process_group()
+
+
+def _run_tree(world):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tree_worker, args=(r, world, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    try:
+        verdict = q.get(timeout=180)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    assert verdict == "ok", verdict
+
+
+def test_tree_plan_fanin_four_ranks_gloo():
+    _run_tree(4)
+
+
+def test_tree_plan_rqc36_shapes():
+    """Tree plans preserve total flops and bound exchanged tensors by the
+    frozen path's own intermediates at every N."""
+    from tnc_amd.dist import make_tree_plan
+    from tnc_amd.fixtures import load_fixture
+
+    for nr in (2, 4, 8):
+        tn, rp, meta = load_fixture("rqc36")
+        plan = make_tree_plan(tn, rp, nr)
+        assert abs(plan.total_flops() - meta["metric_flops"]) < 1e6
+        for ext in plan.externals:
+            assert ext.size() <= meta["peak_size_elems"]
+        assert len(plan.path.toplevel) == plan.nparts - 1

@@ -120,6 +120,133 @@ class DistPlan:
         return sum(self.local_flops(p) for p in range(self.nparts)) + self.fanin_flops
 
 
+def make_tree_plan(tn: CompositeTensor, replace_toplevel, nranks: int) -> DistPlan:
+    """Partition by cutting the (frozen) contraction tree: repeatedly split
+    the heaviest subtree until `nranks` subtrees remain. Each subtree's
+    leaves form one partition contracted locally by the subtree's own steps;
+    the upper tree becomes the fan-in path. Total executed flops equal the
+    single-GPU path's, and every exchanged tensor is an intermediate the
+    1-GPU plan already materializes (so memory is bounded by the frozen
+    path's own peak) — unlike a fresh k-way min-cut, whose boundary tensors
+    on amplitude networks can blow up astronomically.
+
+    The reference reaches feasible plans via KaHyPar + SA instead
+    (repartitioning); this construction is an MI355X-side improvement with
+    identical results. The serial heaviest step still bounds the critical
+    path (the reference has no slicing either; book/src/future_work.md)."""
+    n = len(tn.tensors)
+    assert nranks >= 2
+    # contraction tree from the replace-left path: node = (kind, payload)
+    slot_node = {i: ("leaf", i) for i in range(n)}
+    views = [LeafTensor(t.legs, t.bond_dims) for t in tn.tensors]
+    nodes = []  # internal: dict(left, right, flops_subtree, step_index)
+    node_views = {}
+
+    def node_flops(node):
+        return 0.0 if node[0] == "leaf" else nodes[node[1]]["flops"]
+
+    def view_of(node):
+        return (views[node[1]] if node[0] == "leaf"
+                else nodes[node[1]]["view"])
+
+    for step_idx, (i, j) in enumerate(replace_toplevel):
+        a, b = slot_node[i], slot_node[j]
+        va, vb = view_of(a), view_of(b)
+        out = va ^ vb
+        nodes.append({
+            "left": a, "right": b, "view": out,
+            "flops": node_flops(a) + node_flops(b)
+                     + contract_cost_tensors(va, vb),
+            "step": step_idx,
+        })
+        slot_node[i] = ("node", len(nodes) - 1)
+        slot_node.pop(j)
+    (root,) = slot_node.values()
+
+    # split the heaviest subtree until we have nranks components
+    components = [root]
+    while len(components) < nranks:
+        components.sort(key=node_flops, reverse=True)
+        heavy = components[0]
+        if heavy[0] == "leaf":
+            break  # cannot split further
+        components = components[1:] + [nodes[heavy[1]]["left"],
+                                       nodes[heavy[1]]["right"]]
+    # leaves of each component, in global order
+    def collect(node, out):
+        if node[0] == "leaf":
+            out.append(node[1])
+        else:
+            collect(nodes[node[1]]["left"], out)
+            collect(nodes[node[1]]["right"], out)
+
+    comp_leaves = []
+    for comp in components:
+        ls = []
+        collect(comp, ls)
+        comp_leaves.append(sorted(ls))
+    # partitioning vector; partition ids in component order
+    partitioning = [0] * n
+    for pid, ls in enumerate(comp_leaves):
+        for g in ls:
+            partitioning[g] = pid
+    ptn = partition_tensor_network(tn, partitioning)
+    # partition order in ptn = first appearance of pid; remap to that order
+    order = []
+    for p in partitioning:
+        if p not in order:
+            order.append(p)
+    pos = {p: i for i, p in enumerate(order)}
+
+    # local (nested) paths: original steps whose node lies inside a
+    # component, re-indexed to local leaf positions; fan-in = the rest,
+    # in original step order, over partition positions.
+    comp_of_node = {}
+    for pid, comp in enumerate(components):
+        def mark(node, pid=pid):
+            if node[0] == "node":
+                comp_of_node[node[1]] = pid
+                mark(nodes[node[1]]["left"])
+                mark(nodes[node[1]]["right"])
+        if comp[0] == "node":
+            mark(comp)
+    local_index = {}  # global leaf -> (pid, local idx)
+    for pid, ls in enumerate(comp_leaves):
+        for li, g in enumerate(ls):
+            local_index[g] = (pid, li)
+
+    nested = {}
+    # walk the original path again tracking, per node, its "slot": either
+    # (pid, local slot) while inside a component, or partition position at
+    # and above the cut
+    slot_of = {}
+    for i in range(n):
+        slot_of[("leaf", i)] = ("local", *local_index[i])
+    toplevel = []
+    for idx, nd in enumerate(nodes):
+        a, b = slot_of[nd["left"]], slot_of[nd["right"]]
+        if idx in comp_of_node:
+            pid = comp_of_node[idx]
+            assert a[0] == "local" and b[0] == "local"
+            nested.setdefault(pos[pid], ContractionPath()).toplevel.append(
+                (a[2], b[2]))
+            slot_of[("node", idx)] = a
+        else:
+            # fan-in merge: operands are component results
+            def part_pos(s):
+                if s[0] == "local":
+                    return pos[s[1]]
+                return s[1]
+            pa, pb = part_pos(a), part_pos(b)
+            toplevel.append((pa, pb))
+            slot_of[("node", idx)] = ("part", pa)
+    # single-leaf components appear only via their leaf slots; components
+    # that are whole subtrees already produced their final at their root.
+    # Components never merged (nranks == 1 case) cannot happen here.
+    path = ContractionPath(nested=nested, toplevel=toplevel)
+    return DistPlan(ptn, path, nranks)
+
+
 def make_plan(tn: CompositeTensor, nranks: int, trials: int = 16,
               size_cap=None, seed: int = 0, sa_seconds: float = 0.0) -> DistPlan:
     """Partition + per-partition paths + fan-in path, all deterministic.
