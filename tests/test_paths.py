@@ -167,3 +167,37 @@ def test_optimal_matches_oracle_result():
     out = contract_network(network_to_otensors(tn),
                            Optimal().find_path(tn).replace_path())
     np.testing.assert_allclose(out.data, ref.data, rtol=1e-12)
+
+
+def test_partition_search_beats_rg_on_rqc36():
+    """The HyperOptimizer-substitute quality tier (config 3): on the 36q
+    fixture it finds a substantially cheaper path than RandomGreedy, and the
+    path stays valid and memory-capped."""
+    from tnc_amd import PartitionSearch, RandomGreedy
+    from tnc_amd.fixtures import load_fixture
+
+    tn, rp, meta = load_fixture("rqc36")
+    rg = RandomGreedy(8, size_cap=6.0e9).find_path(tn)
+    ps = PartitionSearch(ks=(2, 4), seeds=(0,), trials=8,
+                         size_cap=6.0e9).find_path(tn)
+    assert ps.flops <= rg.flops
+    assert ps.flops < 0.6 * rg.flops  # the observed win is ~3x
+    validate_path(ps.replace_path())
+    assert ps.size <= 6.0e9
+
+
+def test_partition_search_result_correct():
+    import numpy as np
+
+    from oracle import contract_network
+    from oracle.adapters import network_to_otensors
+    from tnc_amd import PartitionSearch
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+
+    tn = random_circuit(14, 9, 0.5, 0.7, 9, ConnectivityLayout.EAGLE)
+    ref = contract_network(network_to_otensors(tn),
+                           Greedy().find_path(tn).replace_path())
+    ps = PartitionSearch(ks=(2, 3), seeds=(0, 1), trials=4).find_path(tn)
+    out = contract_network(network_to_otensors(tn), ps.replace_path())
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)

@@ -20,7 +20,7 @@ from .cost import (
     contract_path_cost,
     communication_path_cost,
 )
-from .paths import Greedy, Optimal, RandomGreedy, BasicContractionPathResult
+from .paths import Greedy, Optimal, PartitionSearch, RandomGreedy, BasicContractionPathResult
 from .circuit import Circuit, Permutor
 from .connectivity import ConnectivityLayout, connectivity_edges
 from .builders import random_circuit, sycamore_circuit
