@@ -8,7 +8,7 @@ import sys
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-from tnc_amd import RandomGreedy
+from tnc_amd import PartitionSearch, RandomGreedy
 from tnc_amd.cost import contract_cost_tensors
 from tnc_amd.fixtures import FIXTURES, build_fixture, fixture_path, save_network
 from tnc_amd.tensor import LeafTensor
@@ -28,7 +28,12 @@ def metric_flops(tn, replace_toplevel):
 def main():
     for name, cfg in FIXTURES.items():
         tn = build_fixture(name)
-        rg = RandomGreedy(cfg["trials"], size_cap=cfg.get("size_cap")).find_path(tn)
+        # best-of search (the config-3 path is "the best path
+        # found" — RandomGreedy(many) + the partition-guided quality tier
+        # standing in for cotengra HyperOptimizer)
+        rg = PartitionSearch(ks=(2, 3, 4, 6, 8, 12, 16), seeds=(0, 1, 2, 3),
+                             trials=cfg["trials"],
+                             size_cap=cfg.get("size_cap")).find_path(tn)
         replace = rg.replace_path()
         assert not replace.nested
         flops = metric_flops(tn, replace.toplevel)
