@@ -9,7 +9,8 @@ from tnc_amd.contraction_path import ContractionPath
 from tnc_amd.executor import ContractionEngine
 from tnc_amd.fixtures import load_fixture
 
-KIND = {0: "smallk", 1: "dot", 2: "gemm", 3: "gemm+unpack"}
+KIND = {0: "smallk", 1: "dot", 2: "gemm", 3: "gemm+unpack", 5: "dot-lin",
+        6: "dot-tile"}
 
 
 def main():

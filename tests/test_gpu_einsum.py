@@ -193,6 +193,20 @@ def _run_c64(a_labels, a_shape, b_labels, b_shape, seed=0, rtol=2e-3):
     np.testing.assert_allclose(got, ref, rtol=rtol, atol=1e-4)
 
 
+def test_dot_tiled_bitperm():
+    # K = 2^20 with scrambled leg orders -> tiled bit-permutation dot
+    rng = np.random.default_rng(7)
+    legs = list(range(20))
+    blabels = [legs[p] for p in rng.permutation(20)]
+    run_case(legs, [2] * 20, blabels, [2] * 20, rtol=1e-9)
+
+
+def test_dot_tiled_mixed_pow2_dims():
+    # pow2 dims > 2 decompose into bits; K = 2^21 -> tiled path
+    run_case([0, 1, 2, 3, 4, 5, 6], [4, 8, 2, 16, 4, 64, 8],
+             [6, 3, 0, 2, 5, 4, 1], [8, 16, 4, 2, 64, 4, 8], rtol=1e-9)
+
+
 def test_c64_smallk():
     _run_c64([0, 1, 2, 3, 4, 5], [2] * 6, [9, 10, 2, 4], [2] * 4)
 

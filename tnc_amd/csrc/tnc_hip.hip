@@ -23,6 +23,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
@@ -185,6 +186,99 @@ __global__ void k_dot_partial_linear(const CT* __restrict__ A,
     __syncthreads();
   }
   if (threadIdx.x == 0) ws[blockIdx.x] = make_double2(sre[0], sim[0]);
+}
+
+// Tiled dot for a contracted index that is a pure bit-permutation between
+// two contiguous pow2 operands (the final amplitude dot of rqc networks:
+// both operands hold the same 2^30-ish legs in scrambled orders). A naive
+// gather reads one side as random 16B accesses (~1.6 TB/s); here both
+// operands are read coalesced and the scramble happens in LDS.
+//   a-bits: the 6 lowest A bits (lane index, coalesced A reads)
+//   b-bits: the 7 lowest B bits not among the a-bits (coalesced B reads)
+//   rest  : the remaining bits (one block per combination)
+// Each block stages its 128x64 B tile in LDS (XOR-swizzled so both the
+// b-major writes and a-major reads are bank-conflict-free), then streams A.
+#define TN_DOT_TILE_ABITS 6
+#define TN_DOT_TILE_BBITS 7
+#define TN_DOT_TILE_BITS (TN_DOT_TILE_ABITS + TN_DOT_TILE_BBITS)
+#define TN_DOT_MAXBITS 34
+
+struct DotPerm {
+  int rbits;
+  u64 restA[TN_DOT_MAXBITS];  // element strides of each rest bit
+  u64 restB[TN_DOT_MAXBITS];
+  u64 bA[TN_DOT_TILE_BBITS];  // A strides of the b-bits
+  u64 bB[TN_DOT_TILE_BBITS];  // B strides of the b-bits (ascending)
+  u64 aB[TN_DOT_TILE_ABITS];  // B strides of the a-bits (A strides are 1<<i)
+};
+
+template <typename CT>
+__global__ __launch_bounds__(512) void k_dot_tile(const CT* __restrict__ A,
+                                                  const CT* __restrict__ B,
+                                                  double2* __restrict__ ws,
+                                                  DotPerm dp) {
+  __shared__ CT tile[128 * 64];
+  __shared__ u64 sboffA[128], sboffB[128], saoffB[64];
+  __shared__ double sred[16];
+  const int tid = threadIdx.x;
+  if (tid < 128) {
+    u64 oa = 0, ob = 0;
+    for (int i = 0; i < TN_DOT_TILE_BBITS; ++i)
+      if (tid >> i & 1) {
+        oa += dp.bA[i];
+        ob += dp.bB[i];
+      }
+    sboffA[tid] = oa;
+    sboffB[tid] = ob;
+  } else if (tid < 192) {
+    const int a = tid - 128;
+    u64 ob = 0;
+    for (int i = 0; i < TN_DOT_TILE_ABITS; ++i)
+      if (a >> i & 1) ob += dp.aB[i];
+    saoffB[a] = ob;
+  }
+  u64 baseA = 0, baseB = 0;
+  {
+    unsigned r = blockIdx.x;
+    for (int i = 0; i < dp.rbits; ++i) {
+      if (r & 1) {
+        baseA += dp.restA[i];
+        baseB += dp.restB[i];
+      }
+      r >>= 1;
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < 128 * 64; e += 512) {
+    const int b = e & 127, a = e >> 7;
+    tile[b * 64 + (a ^ (b & 63))] = B[baseB + saoffB[a] + sboffB[b]];
+  }
+  __syncthreads();
+  const int wave = tid >> 6, lane = tid & 63;
+  double re = 0.0, im = 0.0;
+  for (int b = wave; b < 128; b += 8) {
+    const CT av = A[baseA + sboffA[b] + (u64)lane];
+    const CT bv = tile[b * 64 + (lane ^ (b & 63))];
+    re = fma((double)av.x, (double)bv.x, fma(-(double)av.y, (double)bv.y, re));
+    im = fma((double)av.x, (double)bv.y, fma((double)av.y, (double)bv.x, im));
+  }
+  for (int s = 32; s > 0; s >>= 1) {
+    re += __shfl_xor(re, s, 64);
+    im += __shfl_xor(im, s, 64);
+  }
+  if (lane == 0) {
+    sred[wave * 2] = re;
+    sred[wave * 2 + 1] = im;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    double tre = 0.0, tim = 0.0;
+    for (int w = 0; w < 8; ++w) {
+      tre += sred[w * 2];
+      tim += sred[w * 2 + 1];
+    }
+    ws[blockIdx.x] = make_double2(tre, tim);
+  }
 }
 
 template <bool P2, typename CT>
@@ -984,6 +1078,76 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
         if (kax[i].sa != pstride || kax[i].sb != pstride) linear = false;
         pstride *= (i64)kax[i].dim;
       }
+    }
+    // tiled bit-permutation path: both operands are contiguous pow2 spans
+    // of the same K elements in different axis orders
+    DotPerm dp;
+    int tbits = -1;
+    if (!linear && K >= (1ull << 20)) {
+      struct BitSB {
+        u64 sa, sb;
+      };
+      std::vector<BitSB> bits;
+      bool ok = true;
+      for (auto& ax : kax) {
+        if ((ax.dim & (ax.dim - 1)) || ax.sa <= 0 || ax.sb <= 0) {
+          ok = false;
+          break;
+        }
+        for (u64 d = 1; d < ax.dim; d <<= 1)
+          bits.push_back({(u64)ax.sa * d, (u64)ax.sb * d});
+      }
+      const int nb = (int)bits.size();
+      if (ok && nb <= TN_DOT_MAXBITS && nb > TN_DOT_TILE_BITS) {
+        std::vector<int> byA(nb), byB(nb);
+        for (int i = 0; i < nb; ++i) byA[i] = byB[i] = i;
+        std::sort(byA.begin(), byA.end(),
+                  [&](int x, int y) { return bits[x].sa < bits[y].sa; });
+        std::sort(byB.begin(), byB.end(),
+                  [&](int x, int y) { return bits[x].sb < bits[y].sb; });
+        for (int i = 0; i < nb; ++i)
+          if (bits[byA[i]].sa != (1ull << i) || bits[byB[i]].sb != (1ull << i))
+            ok = false;  // not a contiguous span on both sides
+        if (ok) {
+          std::vector<char> used(nb, 0);
+          for (int i = 0; i < TN_DOT_TILE_ABITS; ++i) {
+            used[byA[i]] = 1;
+            dp.aB[i] = bits[byA[i]].sb;
+          }
+          int nbb = 0, nr = 0;
+          for (int i = 0; i < nb && nbb < TN_DOT_TILE_BBITS; ++i) {
+            const int bi = byB[i];
+            if (used[bi]) continue;
+            dp.bA[nbb] = bits[bi].sa;
+            dp.bB[nbb] = bits[bi].sb;
+            used[bi] = 1;
+            ++nbb;
+          }
+          for (int i = 0; i < nb; ++i) {
+            if (used[byA[i]]) continue;
+            dp.restA[nr] = bits[byA[i]].sa;
+            dp.restB[nr] = bits[byA[i]].sb;
+            ++nr;
+          }
+          dp.rbits = nr;
+          tbits = nr;
+        }
+      }
+    }
+    if (tbits > 0) {
+      if (stats) stats->kind = 6;  // tiled bit-permutation dot
+      const u64 nblk = 1ull << tbits;
+      double2* wsbuf;
+      {
+        int rc_ = ws_alloc(ws, (void**)&wsbuf, nblk * sizeof(double2));
+        if (rc_) return rc_;
+      }
+      k_dot_tile<<<dim3((unsigned)nblk), 512, 0, stream>>>(Adata, Bdata,
+                                                           wsbuf, dp);
+      k_dot_finish<<<1, 256, 0, stream>>>(wsbuf, out, (int)nblk);
+      ws_free(ws, wsbuf);
+      HIP_CHECK(hipGetLastError());
+      return TN_OK;
     }
     if (stats) stats->kind = linear ? 5 : 1;  // 5 = linear (streaming) dot
     int blocks = grid_for(K);
