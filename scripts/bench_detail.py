@@ -16,7 +16,8 @@ KIND = {0: "smallk", 1: "dot", 2: "gemm", 3: "gemm+unpack", 5: "dot-lin",
 def main():
     name = sys.argv[1] if len(sys.argv) > 1 else "rqc36"
     tn, rp, meta = load_fixture(name)
-    eng = ContractionEngine(tn, ContractionPath.simple(rp))
+    eng = ContractionEngine(tn, ContractionPath.simple(rp),
+                            dtype=meta.get("dtype", "c128"))
     eng.contract()  # warmup
     elapsed, step_ms, gemm_ms, kinds = eng.contract_profiled()
     total = sum(step_ms)

@@ -891,7 +891,17 @@ struct Arena {
 
   int reserve(size_t bytes) {
     if (base) return 0;
-    if (hipMalloc((void**)&base, bytes) != hipSuccess) return -1;
+    // never try to reserve more than the device can hold: clamp to 90% of
+    // free memory so per-step spill allocations still have headroom
+    size_t free_b = 0, total_b = 0;
+    if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && free_b > 0) {
+      size_t cap = free_b - free_b / 10;
+      if (bytes > cap) bytes = cap;
+    }
+    if (hipMalloc((void**)&base, bytes) != hipSuccess) {
+      (void)hipGetLastError();  // swallow the sticky OOM; caller recovers
+      return -1;
+    }
     size = bytes;
     blocks = {{0, bytes, true}};
     return 0;
@@ -959,7 +969,14 @@ static int ws_alloc(WsCtx& ctx, void** p, size_t bytes) {
     if (*p) return TN_OK;
   }
   if (hipMalloc(p, bytes) != hipSuccess) {
-    g_last_error = "device allocation failed";
+    (void)hipGetLastError();  // don't leave a sticky OOM for launch checks
+    size_t fb = 0, tb = 0;
+    (void)hipMemGetInfo(&fb, &tb);
+    char buf[192];
+    snprintf(buf, sizeof buf,
+             "device allocation failed (%zu bytes; %zu free of %zu; arena %s)",
+             bytes, fb, tb, ctx.arena ? "exhausted" : "absent");
+    g_last_error = buf;
     return TN_ERR_OOM;
   }
   return TN_OK;
@@ -1176,7 +1193,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   // MFMA GEMM than as an address-gather stream
   bool skinny = (M < 16 || N < 16);
   bool gemm_worthy = (K >= 16 && M >= MF_T && N >= MF_TN);
-  if ((K <= TN_SMALLK || skinny) && !gemm_worthy) {
+  bool gather_ok = (K <= TN_SMALLK || skinny);
+  auto run_gather = [&]() -> int {
+    if (stats) stats->kind = 0;
     // out map: every out axis, with its source stride in A or B
     std::vector<AxisInfo> oax;
     for (int i = 0; i < out_nd; ++i) {
@@ -1212,7 +1231,8 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     }
     HIP_CHECK(hipGetLastError());
     return TN_OK;
-  }
+  };
+  if (gather_ok && !gemm_worthy) return run_gather();
 
   // ---- TTGT: pack (if needed) + GEMM + unpack (if needed) ----
   if (stats) stats->kind = 2;
@@ -1249,6 +1269,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     u64 elems = M * K;
     {
       int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(CT));
+      // pack workspace doesn't fit -> run the shape through the (slower)
+      // gather kernels instead of failing the whole contraction
+      if (rc_ == TN_ERR_OOM && gather_ok) return run_gather();
       if (rc_) return rc_;
     }
     int blocks = grid_for(elems);
@@ -1267,6 +1290,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     u64 elems = K * N;
     {
       int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(CT));
+      if (rc_ == TN_ERR_OOM && gather_ok) {
+        ws_free(ws, packA);
+        return run_gather();
+      }
       if (rc_) return rc_;
     }
     int blocks = grid_for(elems);
@@ -1288,6 +1315,11 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if (stats) stats->kind = 3;
     {
       int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(CT));
+      if (rc_ == TN_ERR_OOM && gather_ok) {
+        ws_free(ws, packA);
+        ws_free(ws, packB);
+        return run_gather();
+      }
       if (rc_) return rc_;
     }
     Cg = tmpC;
@@ -1313,8 +1345,14 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   CT* splitbuf = nullptr;
   if (splitk > 1) {
     int rc_ = ws_alloc(ws, (void**)&splitbuf, splitk * nout * sizeof(CT));
-    if (rc_) return rc_;
-    gemm_out = splitbuf;
+    if (rc_ == TN_ERR_OOM) {
+      splitk = 1;  // split-K is an optimization; run unsplit when tight
+      kchunk = ((K + MF_K - 1) / MF_K) * MF_K;
+    } else if (rc_) {
+      return rc_;
+    } else {
+      gemm_out = splitbuf;
+    }
   }
   dim3 grid((unsigned)(tiles * splitk));
   if (stats && stats->gemm_ev0)
@@ -1790,6 +1828,10 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     if (kind) kind[s] = st.kind;
     if (profiled) kinds_local[s] = st.kind;
     if (rc != TN_OK) {
+      char buf[96];
+      snprintf(buf, sizeof buf, " (at step %zu, out elems %llu)", s,
+               (unsigned long long)out.elems);
+      g_last_error += buf;
       ws_free(ws, out.data);
       break;
     }

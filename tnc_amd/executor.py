@@ -18,19 +18,25 @@ from .tensor import CompositeTensor, LeafTensor, TensorData
 
 
 class StepInfo:
-    __slots__ = ("i", "j", "m", "n", "k", "flops", "out_legs", "out_dims")
+    __slots__ = ("i", "j", "m", "n", "k", "flops", "out_legs", "out_dims",
+                 "packa", "packb")
 
-    def __init__(self, i, j, m, n, k, flops, out_legs, out_dims):
+    def __init__(self, i, j, m, n, k, flops, out_legs, out_dims,
+                 packa=False, packb=False):
         self.i, self.j = i, j
         self.m, self.n, self.k = m, n, k
         self.flops = flops
         self.out_legs = out_legs
         self.out_dims = out_dims
+        self.packa = packa  # TTGT would permute A into [M..][K..] order
+        self.packb = packb  # TTGT would permute B into [K..][N..] order
 
 
 def plan_steps(leaves, steps):
     """Host-side metadata walk: per-step (M, N, K), metric flops
-    ((8s-2)*o, contraction_cost.rs:26-32) and output legs."""
+    ((8s-2)*o, contraction_cost.rs:26-32), output legs, and whether the
+    TTGT path would need pack permutes (mirrors einsum_dev_impl's
+    is_ready checks in tnc_amd/csrc/tnc_hip.hip)."""
     views = [LeafTensor(t.legs, t.bond_dims) for t in leaves]
     infos = []
     for i, j in steps:
@@ -40,24 +46,50 @@ def plan_steps(leaves, steps):
         m = (a - b).size()
         n = (b - a).size()
         k = shared.size()
+        aset, bset = set(a.legs), set(b.legs)
+        shared_a = [l for l in a.legs if l in bset]  # K legs in A order
+        a_axes = [x for x, l in enumerate(a.legs) if l not in bset]
+        a_axes += [a.legs.index(l) for l in shared_a]
+        b_axes = [b.legs.index(l) for l in shared_a]
+        b_axes += [x for x, l in enumerate(b.legs) if l not in aset]
         infos.append(
-            StepInfo(i, j, m, n, k, contract_cost_tensors(a, b), out.legs, out.bond_dims)
+            StepInfo(i, j, m, n, k, contract_cost_tensors(a, b), out.legs,
+                     out.bond_dims,
+                     packa=a_axes != list(range(len(a.legs))),
+                     packb=b_axes != list(range(len(b.legs))))
         )
         views[i] = out
         views[j] = None
     return infos
 
 
+def _step_ws_bytes(info, esize):
+    """Workspace the C dispatch will allocate for one step (mirrors
+    einsum_dev_impl: dot partials, nothing for gather kernels, pack
+    buffers for TTGT). Keep in sync with TN_SMALLK=64, MF_T=128,
+    MF_TN=64 in tnc_amd/csrc/tnc_hip.hip."""
+    if info.m == 1 and info.n == 1 and info.k > 64:
+        return 1 << 25  # dot partial buffer (<= 2^21 blocks * 16 B)
+    gemm_worthy = info.k >= 16 and info.m >= 128 and info.n >= 64
+    if (info.k <= 64 or info.m < 16 or info.n < 16) and not gemm_worthy:
+        return 0  # smallk/anyk gather: no workspace
+    ws = 0
+    if info.packa:
+        ws += info.m * info.k * esize
+    if info.packb:
+        ws += info.k * info.n * esize
+    return ws
+
+
 def arena_bytes(leaves, steps, infos, esize=16):
     """Peak device-arena demand: live intermediates + this step's output and
-    worst-case packing workspaces (A' + B' + unpacked C), walked over the
-    plan. `esize` bytes per element; padded 15% for fragmentation."""
+    the workspaces its dispatch actually uses, walked over the plan.
+    Padded 15% for fragmentation + split-K slack."""
     live = {}  # slot -> bytes (intermediates only; leaves live outside)
     peak = 0
     for info in infos:
         out_b = info.m * info.n * esize
-        extra = (info.m * info.k + info.k * info.n) * esize + out_b
-        demand = sum(live.values()) + out_b + extra
+        demand = sum(live.values()) + out_b + _step_ws_bytes(info, esize)
         peak = max(peak, demand)
         live.pop(info.j, None)
         live[info.i] = out_b
