@@ -68,7 +68,9 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds, dtype="c128",
                 pmc = json.load(f)
             entry = pmc.get("fixtures", {}).get(fixture)
             if entry:
-                traffic = entry.get("dominant_kernel_traffic_bytes")
+                key = f"{int(info.m)}x{int(info.n)}x{int(info.k)}"
+                traffic = entry.get("by_mnk", {}).get(
+                    key, entry.get("dominant_kernel_traffic_bytes"))
         except Exception:
             traffic = None
     mfma_peak = F64_MFMA_PEAK if dtype == "c128" else F32_MFMA_PEAK

@@ -1855,7 +1855,9 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
         step_ms[s] = (double)sms;
         if (gemm_ms) {
           float gms = 0.f;
-          if (kinds_local[s] >= 2 &&
+          // only TTGT steps (kind 2/3) record the gemm events; probing
+          // unrecorded events would fail AND leave a sticky TLS error
+          if ((kinds_local[s] == 2 || kinds_local[s] == 3) &&
               hipEventElapsedTime(&gms, ev[4 * s + 2], ev[4 * s + 3]) ==
                   hipSuccess)
             gemm_ms[s] = (double)gms;
@@ -1863,6 +1865,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
             gemm_ms[s] = 0.0;
         }
       }
+      (void)hipGetLastError();  // clear any error from unrecorded-event reads
     }
     // locate final tensor
     int final_idx = -1;
