@@ -85,7 +85,8 @@ def roofline_from_profile(infos, step_ms, gemm_ms, kinds, dtype="c128",
             "unit": "FLOP/s",
             "frac": achieved / mfma_peak,
             "traffic": traffic,
-            "kernel": "k_zgemm_mfma",
+            "kernel": ("k_zgemm_c128_glds_pure" if dtype == "c128"
+                       else "k_zgemm_c64_glds_pure"),
             "launch_ms": gemm_ms[dom],
             "mnk": [info.m, info.n, info.k],
         }
