@@ -250,6 +250,9 @@ def run_distributed(args):
     device = torch.device(f"cuda:{local_rank}")
 
     tn, frozen_path, meta = load_fixture(args.fixture)
+    dtype = meta.get("dtype", "c128")
+    esize = 16 if dtype == "c128" else 8
+    torch_view = torch.float64 if dtype == "c128" else torch.float32
     if frozen_path:
         # cut the frozen contraction tree into `world` subtrees: total flops
         # preserved, every exchanged tensor is a path intermediate (bounded
@@ -268,11 +271,11 @@ def run_distributed(args):
         sub = plan.partitioned.tensors[my_part]
         if isinstance(sub, CompositeTensor):
             inner = plan.path.nested.get(my_part, ContractionPath.simple([]))
-            eng = ContractionEngine(sub, inner, device=local_rank)
+            eng = ContractionEngine(sub, inner, device=local_rank, dtype=dtype)
         else:
             eng = ContractionEngine(CompositeTensor([sub]),
                                     ContractionPath.simple([]),
-                                    device=local_rank)
+                                    device=local_rank, dtype=dtype)
 
     L = hiplib.lib()
     import math as _math
@@ -290,20 +293,20 @@ def run_distributed(args):
 
         def send(self, handle, legs, dims, peer):
             elems = int(_math.prod(dims)) if dims else 1
-            t = torch.empty((elems, 2), dtype=torch.float64, device=device)
+            t = torch.empty((elems, 2), dtype=torch_view, device=device)
             hiplib.check(L.tn_memcpy_dtod(t.data_ptr(), self.ptr_of(handle),
-                                          elems * 16), "tn_memcpy_dtod")
+                                          elems * esize), "tn_memcpy_dtod")
             dist_t.send(t, dst=peer)
 
         def recv(self, legs, dims, peer):
             elems = int(_math.prod(dims)) if dims else 1
-            t = torch.empty((elems, 2), dtype=torch.float64, device=device)
+            t = torch.empty((elems, 2), dtype=torch_view, device=device)
             dist_t.recv(t, src=peer)
             self.keep.append(t)
             return ("torch", t)
 
         def contract_pair(self, a, a_legs, a_dims, b, b_legs, b_dims):
-            net = L.tn_net_create(local_rank)
+            net = L.tn_net_create2(local_rank, 0 if dtype == "c128" else 1)
             if not net:
                 raise RuntimeError(hiplib.last_error())
             self.nets.append(net)
@@ -352,7 +355,7 @@ def run_distributed(args):
         flops = plan.total_flops()
         value = flops * args.steps / wall / 1e9
         emit({
-            "metric": "pairwise-contraction GFLOP/s (c128)",
+            "metric": f"pairwise-contraction GFLOP/s ({dtype})",
             "value": value,
             "unit": "GFLOP/s",
             "n_gpus": world,
@@ -362,7 +365,7 @@ def run_distributed(args):
             "higher_is_better": True,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "c128",
+            "dtype": dtype,
             "data": "synthetic",
             "config": {
                 "workload": f"{args.fixture}: 36q depth-14 RQC amplitude, "
