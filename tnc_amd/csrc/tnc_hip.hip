@@ -1004,6 +1004,12 @@ struct Arena {
 struct WsCtx {
   Arena* arena = nullptr;  // optional
   hipStream_t stream = nullptr;
+  // stream-capture mode (hipGraph): non-arena allocations poison the graph
+  // (their pointers would dangle on replay) and frees must be deferred past
+  // EndCapture (hipFree device-syncs, which is illegal mid-capture)
+  bool capturing = false;
+  bool spilled = false;
+  std::vector<void*>* deferred = nullptr;
 };
 
 // Non-arena workspace uses plain hipMalloc, NOT hipMallocAsync: on this
@@ -1017,6 +1023,7 @@ static int ws_alloc(WsCtx& ctx, void** p, size_t bytes) {
     *p = ctx.arena->alloc(bytes);
     if (*p) return TN_OK;
   }
+  if (ctx.capturing) ctx.spilled = true;
   if (hipMalloc(p, bytes) != hipSuccess) {
     (void)hipGetLastError();  // don't leave a sticky OOM for launch checks
     size_t fb = 0, tb = 0;
@@ -1035,6 +1042,10 @@ static void ws_free(WsCtx& ctx, void* p) {
   if (!p) return;
   if (ctx.arena && ctx.arena->owns(p)) {
     ctx.arena->release(p);
+    return;
+  }
+  if (ctx.capturing && ctx.deferred) {
+    ctx.deferred->push_back(p);
     return;
   }
   (void)hipFree(p);  // hipFree device-syncs before releasing the pages
@@ -1674,6 +1685,24 @@ struct tn_net {
   bool has_final = false;
   Arena arena;
   u64 pool_in_use = 0;
+  // hipGraph replay of the (launch-bound) walk: captured after the first
+  // normal run when every workspace block came from the arena, so the
+  // graph's baked device pointers are stable across replays (the arena's
+  // host-side first-fit is deterministic and the final block stays
+  // reserved). Invalidated by a different path or new leaves.
+  hipGraphExec_t graph_exec = nullptr;
+  std::vector<u64> graph_pairs;
+  int graph_state = 0;  // 0 = no normal run yet, 1 = ready to capture,
+                        // 2 = captured, -1 = disabled (spill/failure)
+  DevTensor graph_final;  // meta + data ptr of the captured final (not owned)
+  bool graph_final_in_arena = false;
+
+  void invalidate_graph() {
+    if (graph_exec) (void)hipGraphExecDestroy(graph_exec);
+    graph_exec = nullptr;
+    graph_pairs.clear();
+    if (graph_state > 0) graph_state = 0;
+  }
 };
 
 extern "C" tn_net* tn_net_create2(int device, int dtype) {
@@ -1708,6 +1737,7 @@ extern "C" tn_net* tn_net_create(int device) {
 
 extern "C" int tn_net_reserve(tn_net* net, uint64_t bytes) {
   if (!net) FAILV(TN_ERR_INVALID, "null net");
+  net->invalidate_graph();
   HIP_CHECK(hipSetDevice(net->device));
   if (net->arena.reserve(bytes))
     FAILV(TN_ERR_OOM, "arena reservation of %llu bytes failed",
@@ -1723,6 +1753,7 @@ extern "C" int64_t tn_net_add_leaf(tn_net* net, const u64* labels,
     g_last_error = "rank too large";
     return -TN_ERR_INVALID;
   }
+  net->invalidate_graph();  // the network changed
   if (hipSetDevice(net->device) != hipSuccess) return -TN_ERR_HIP;
   DevTensor t;
   t.labels.assign(labels, labels + ndim);
@@ -1751,6 +1782,7 @@ extern "C" int64_t tn_net_add_leaf_dev(tn_net* net, const u64* labels,
     g_last_error = "rank too large";
     return -TN_ERR_INVALID;
   }
+  net->invalidate_graph();  // the network changed
   DevTensor t;
   t.labels.assign(labels, labels + ndim);
   t.dims.assign(dims, dims + ndim);
@@ -1788,10 +1820,13 @@ static void symdiff(const DevTensor& a, const DevTensor& b,
 
 static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
                          double* step_ms, double* gemm_ms, int32_t* kind,
-                         double* elapsed_ms) {
+                         double* elapsed_ms, bool capture = false,
+                         std::vector<void*>* deferred = nullptr,
+                         bool* spilled = nullptr) {
   if (!net) FAILV(TN_ERR_INVALID, "null net");
   HIP_CHECK(hipSetDevice(net->device));
-  WsCtx ws{net->arena.base ? &net->arena : nullptr, net->stream};
+  WsCtx ws{net->arena.base ? &net->arena : nullptr, net->stream, capture,
+           false, deferred};
   if (net->has_final && net->final_t.owned) {
     if (net->final_in_arena)
       net->arena.release(net->final_t.data);
@@ -1815,12 +1850,14 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     kinds_local.assign(nsteps, 0);
   }
 
-  HIP_CHECK(hipStreamSynchronize(net->stream));
   double t0_ms = 0.0;
   hipEvent_t walk_start = nullptr, walk_end = nullptr;
-  HIP_CHECK(hipEventCreate(&walk_start));
-  HIP_CHECK(hipEventCreate(&walk_end));
-  HIP_CHECK(hipEventRecord(walk_start, net->stream));
+  if (!capture) {
+    HIP_CHECK(hipStreamSynchronize(net->stream));
+    HIP_CHECK(hipEventCreate(&walk_start));
+    HIP_CHECK(hipEventCreate(&walk_end));
+    HIP_CHECK(hipEventRecord(walk_start, net->stream));
+  }
 
   int rc = TN_OK;
   for (size_t s = 0; s < nsteps; ++s) {
@@ -1898,11 +1935,15 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     alive[j] = 0;
   }
 
-  HIP_CHECK(hipEventRecord(walk_end, net->stream));
-  HIP_CHECK(hipStreamSynchronize(net->stream));
+  if (!capture) {
+    HIP_CHECK(hipEventRecord(walk_end, net->stream));
+    HIP_CHECK(hipStreamSynchronize(net->stream));
+  }
   if (rc == TN_OK) {
     float ms = 0.f;
-    HIP_CHECK(hipEventElapsedTime(&ms, walk_start, walk_end));
+    if (!capture) {
+      HIP_CHECK(hipEventElapsedTime(&ms, walk_start, walk_end));
+    }
     if (elapsed_ms) *elapsed_ms = (double)ms + t0_ms;
     if (profiled) {
       for (size_t s = 0; s < nsteps; ++s) {
@@ -1961,25 +2002,135 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     for (size_t x = 0; x < n; ++x)
       if (alive[x] && slots[x].owned && slots[x].data)
         ws_free(ws, slots[x].data);
-    (void)hipStreamSynchronize(net->stream);
+    if (!capture) (void)hipStreamSynchronize(net->stream);
   }
-  (void)hipEventDestroy(walk_start);
-  (void)hipEventDestroy(walk_end);
+  if (spilled) *spilled = ws.spilled;
+  if (walk_start) (void)hipEventDestroy(walk_start);
+  if (walk_end) (void)hipEventDestroy(walk_end);
   for (auto& e : ev) (void)hipEventDestroy(e);
   return rc;
 }
 
+static bool graph_pairs_match(const tn_net* net, const u64* pairs,
+                              size_t nsteps) {
+  return net->graph_pairs.size() == 2 * nsteps &&
+         std::equal(net->graph_pairs.begin(), net->graph_pairs.end(), pairs);
+}
+
+static int contract_graph_replay(tn_net* net, double* elapsed_ms) {
+  HIP_CHECK(hipSetDevice(net->device));
+  hipEvent_t e0 = nullptr, e1 = nullptr;
+  HIP_CHECK(hipEventCreate(&e0));
+  HIP_CHECK(hipEventCreate(&e1));
+  HIP_CHECK(hipEventRecord(e0, net->stream));
+  if (hipGraphLaunch(net->graph_exec, net->stream) != hipSuccess) {
+    (void)hipGetLastError();
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    net->invalidate_graph();
+    net->graph_state = -1;
+    FAILV(TN_ERR_HIP, "hipGraphLaunch failed");
+  }
+  HIP_CHECK(hipEventRecord(e1, net->stream));
+  HIP_CHECK(hipStreamSynchronize(net->stream));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+  (void)hipEventDestroy(e0);
+  (void)hipEventDestroy(e1);
+  if (elapsed_ms) *elapsed_ms = (double)ms;
+  // bookkeeping is untouched: the replay rewrites the same arena blocks and
+  // the final tensor lands at its captured address (net->final_t)
+  net->has_final = true;
+  return TN_OK;
+}
+
+static int contract_graph_capture(tn_net* net, const u64* pairs,
+                                  size_t nsteps) {
+  HIP_CHECK(hipSetDevice(net->device));
+  // release the previous final OUTSIDE capture (the release may hipFree,
+  // and hipFree device-syncs — illegal while the stream captures)
+  if (net->has_final && net->final_t.owned) {
+    if (net->final_in_arena)
+      net->arena.release(net->final_t.data);
+    else
+      (void)hipFree(net->final_t.data);
+    net->final_t = DevTensor();
+    net->has_final = false;
+    net->final_in_arena = false;
+  }
+  HIP_CHECK(hipStreamSynchronize(net->stream));
+  if (hipStreamBeginCapture(net->stream, hipStreamCaptureModeThreadLocal) !=
+      hipSuccess) {
+    (void)hipGetLastError();
+    net->graph_state = -1;
+    return TN_ERR_HIP;
+  }
+  std::vector<void*> deferred;
+  bool spilled = false;
+  int rc = contract_impl(net, pairs, nsteps, nullptr, nullptr, nullptr,
+                         nullptr, /*capture=*/true, &deferred, &spilled);
+  hipGraph_t g = nullptr;
+  hipError_t ce = hipStreamEndCapture(net->stream, &g);
+  for (void* p : deferred) (void)hipFree(p);
+  (void)hipGetLastError();
+  if (rc != TN_OK || ce != hipSuccess || !g || spilled) {
+    if (g) (void)hipGraphDestroy(g);
+    net->graph_state = -1;  // caller falls back to a normal run
+    return rc != TN_OK ? rc : TN_ERR_HIP;
+  }
+  hipGraphExec_t exec = nullptr;
+  if (hipGraphInstantiate(&exec, g, nullptr, nullptr, 0) != hipSuccess) {
+    (void)hipGetLastError();
+    (void)hipGraphDestroy(g);
+    net->graph_state = -1;
+    return TN_ERR_HIP;
+  }
+  (void)hipGraphDestroy(g);
+  net->graph_exec = exec;
+  net->graph_final = net->final_t;  // meta + address of the captured final
+  net->graph_final.owned = false;
+  net->graph_final_in_arena = net->final_in_arena;
+  net->graph_state = 2;
+  return TN_OK;
+}
+
 extern "C" int tn_net_contract(tn_net* net, const u64* pairs, size_t nsteps,
                                double* elapsed_ms) {
-  return contract_impl(net, pairs, nsteps, nullptr, nullptr, nullptr,
-                       elapsed_ms);
+  if (!net) FAILV(TN_ERR_INVALID, "null net");
+  if (net->graph_state == 2) {
+    if (graph_pairs_match(net, pairs, nsteps))
+      return contract_graph_replay(net, elapsed_ms);
+    net->invalidate_graph();  // different path: baked pointers are stale
+  }
+  if (net->graph_state == 1 && net->arena.base &&
+      graph_pairs_match(net, pairs, nsteps)) {
+    if (contract_graph_capture(net, pairs, nsteps) == TN_OK)
+      return contract_graph_replay(net, elapsed_ms);
+    // capture failed (workspace spill etc.) — run normally below
+  }
+  int rc = contract_impl(net, pairs, nsteps, nullptr, nullptr, nullptr,
+                         elapsed_ms);
+  if (rc == TN_OK && net->graph_state >= 0 && net->arena.base) {
+    net->graph_pairs.assign(pairs, pairs + 2 * nsteps);
+    net->graph_state = 1;  // arm capture for the next identical call
+  }
+  return rc;
 }
 
 extern "C" int tn_net_contract_profiled(tn_net* net, const u64* pairs,
                                         size_t nsteps, double* step_ms,
                                         double* gemm_ms, int32_t* kind,
                                         double* elapsed_ms) {
-  return contract_impl(net, pairs, nsteps, step_ms, gemm_ms, kind, elapsed_ms);
+  if (!net) FAILV(TN_ERR_INVALID, "null net");
+  if (net->graph_state == 2 && !graph_pairs_match(net, pairs, nsteps))
+    net->invalidate_graph();  // different path moves the final block
+  int rc =
+      contract_impl(net, pairs, nsteps, step_ms, gemm_ms, kind, elapsed_ms);
+  if (rc == TN_OK && net->graph_state == 0 && net->arena.base) {
+    net->graph_pairs.assign(pairs, pairs + 2 * nsteps);
+    net->graph_state = 1;  // a profiled pass also arms graph capture
+  }
+  return rc;
 }
 
 extern "C" int tn_memcpy_dtod(void* dst, const void* src, u64 bytes) {
@@ -2031,6 +2182,7 @@ extern "C" void tn_net_destroy(tn_net* net) {
   if (!net) return;
   (void)hipSetDevice(net->device);
   (void)hipStreamSynchronize(net->stream);
+  net->invalidate_graph();
   for (auto& t : net->leaves)
     if (t.data && !t.owned && !t.external) (void)hipFree(t.data);
   if (net->has_final && net->final_t.owned && !net->final_in_arena)
