@@ -168,3 +168,36 @@ def test_syc49_engine_builds():
                                    rtol=2e-3, atol=1e-4)
         slots[i] = ref
         slots[j] = None
+
+
+def test_graph_replay_identical():
+    """contract() three times: normal run, hipGraph capture+replay, pure
+    replay — every pass must match the oracle (and each other bit-exactly).
+    Uses the rqc24 fixture so the engine reserves an arena (graph capture
+    only arms with one)."""
+    from tnc_amd.contraction_path import ContractionPath
+    from tnc_amd.executor import ContractionEngine
+    from tnc_amd.fixtures import load_fixture
+
+    from tnc_amd import hiplib
+
+    tn, rp, meta = load_fixture("rqc24")
+    replace = ContractionPath.simple(rp)
+    ref = contract_network(network_to_otensors(tn), replace)
+    eng = ContractionEngine(tn, replace)
+    # rqc24 is below the executor's auto-reserve threshold; force an arena
+    # so graph capture arms (it needs stable workspace addresses)
+    hiplib.check(hiplib.lib().tn_net_reserve(eng.net, 256 * 1024 * 1024),
+                 "tn_net_reserve")
+    try:
+        results = []
+        for _ in range(3):
+            eng.contract()
+            _, data = eng.result()
+            results.append(np.array(data, copy=True))
+        for r in results:
+            np.testing.assert_allclose(r, ref.data, rtol=1e-10, atol=1e-12)
+        assert (results[0] == results[1]).all()
+        assert (results[1] == results[2]).all()
+    finally:
+        eng.close()

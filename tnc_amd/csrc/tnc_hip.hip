@@ -1694,8 +1694,6 @@ struct tn_net {
   std::vector<u64> graph_pairs;
   int graph_state = 0;  // 0 = no normal run yet, 1 = ready to capture,
                         // 2 = captured, -1 = disabled (spill/failure)
-  DevTensor graph_final;  // meta + data ptr of the captured final (not owned)
-  bool graph_final_in_arena = false;
 
   void invalidate_graph() {
     if (graph_exec) (void)hipGraphExecDestroy(graph_exec);
@@ -2087,9 +2085,6 @@ static int contract_graph_capture(tn_net* net, const u64* pairs,
   }
   (void)hipGraphDestroy(g);
   net->graph_exec = exec;
-  net->graph_final = net->final_t;  // meta + address of the captured final
-  net->graph_final.owned = false;
-  net->graph_final_in_arena = net->final_in_arena;
   net->graph_state = 2;
   return TN_OK;
 }
