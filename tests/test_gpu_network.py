@@ -201,3 +201,28 @@ def test_graph_replay_identical():
         assert (results[1] == results[2]).all()
     finally:
         eng.close()
+
+
+def test_qasm_derived_network_gpu():
+    """North-star parity sentence made literal: a QASM-derived amplitude
+    network contracts on the GPU to the oracle's value within 1e-10
+    relative (c128)."""
+    from tnc_amd import RandomGreedy
+    from tnc_amd.qasm import import_qasm
+
+    lines = ['OPENQASM 2.0;', 'include "qelib1.inc";', "qreg q[10];"]
+    for i in range(10):
+        lines.append(f"h q[{i}];")
+    for i in range(9):
+        lines.append(f"cx q[{i}],q[{i+1}];")
+    for i in range(10):
+        lines.append(f"u3(0.3,{0.1*i},-0.2) q[{i}];")
+    for i in range(0, 9, 2):
+        lines.append(f"cp(pi/{i+2}) q[{i}],q[{i+1}];")
+    for i in range(10):
+        lines.append(f"t q[{i}];")
+    lines.append("swap q[0],q[9];")
+    circuit = import_qasm("\n".join(lines))
+    tn, _ = circuit.into_amplitude_network("0" * 10)
+    legs, data, ref = gpu_contract(tn, RandomGreedy(8))
+    np.testing.assert_allclose(data, ref.data, rtol=1e-10)
