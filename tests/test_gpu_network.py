@@ -226,3 +226,27 @@ def test_qasm_derived_network_gpu():
     tn, _ = circuit.into_amplitude_network("0" * 10)
     legs, data, ref = gpu_contract(tn, RandomGreedy(8))
     np.testing.assert_allclose(data, ref.data, rtol=1e-10)
+
+
+def test_sliced_contraction_gpu():
+    """Slicing (memory-for-flops): sum of GPU-contracted slices equals the
+    direct contraction and the oracle."""
+    from tnc_amd import Greedy
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.executor import contract_tensor_network_gpu
+    from tnc_amd.slicing import contract_sliced_gpu, find_slice_edges, \
+        _walk_sizes
+
+    tn = random_circuit(16, 14, 0.5, 0.8, 3, ConnectivityLayout.EAGLE)
+    replace = Greedy().find_path(tn).replace_path()
+    peak, _, _ = _walk_sizes(tn, replace.toplevel)
+    edges, new_peak = find_slice_edges(tn, replace.toplevel, peak / 4,
+                                       max_edges=4)
+    assert edges and new_peak < peak
+    legs_s, sliced = contract_sliced_gpu(tn, replace, edges)
+    legs_d, direct = contract_tensor_network_gpu(tn, replace)
+    ref = contract_network(network_to_otensors(tn), replace)
+    assert legs_s == legs_d == ref.legs
+    np.testing.assert_allclose(sliced, direct, rtol=1e-12, atol=1e-13)
+    np.testing.assert_allclose(sliced, ref.data, rtol=1e-10, atol=1e-12)

@@ -32,3 +32,4 @@ from .repartition import (
     balance_partitions,
     compute_solution,
 )
+from .slicing import find_slice_edges, slice_network
