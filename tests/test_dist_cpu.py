@@ -227,3 +227,68 @@ def test_tree_plan_rqc36_shapes():
         for ext in plan.externals:
             assert ext.size() <= meta["peak_size_elems"]
         assert len(plan.path.toplevel) == plan.nparts - 1
+
+
+def _slice_worker(rank, world, result_queue):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29519"
+    dist_t.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import math
+
+        from oracle import contract_network
+        from oracle.adapters import network_to_otensors
+        from tnc_amd.contraction_path import ContractionPath
+        from tnc_amd.fixtures import load_fixture
+        from tnc_amd.slicing import (find_slice_edges, iter_assignments,
+                                     slice_network)
+
+        tn, rp, meta = load_fixture("rqc24")
+        replace = ContractionPath.simple(rp)
+        ebits = max(1, int(math.ceil(math.log2(world))))
+        edges, _ = find_slice_edges(tn, rp, 0, max_edges=ebits)
+        assignments = list(iter_assignments(tn, edges))
+        assert len(edges) == ebits, (edges, ebits)
+        # the bench's rank->slice mapping: rank r takes assignments[r::world]
+        local = None
+        for a in assignments[rank::world]:
+            stn = slice_network(tn, a)
+            part = contract_network(network_to_otensors(stn), replace)
+            arr = np.atleast_1d(np.asarray(part.data, dtype=np.complex128))
+            local = arr.copy() if local is None else local + arr
+        if local is None:
+            local = np.zeros(1, dtype=np.complex128)
+        buf = torch.from_numpy(local.view(np.float64))
+        dist_t.all_reduce(buf, op=dist_t.ReduceOp.SUM)
+        total = buf.numpy().view(np.complex128)
+        if rank == 0:
+            direct = contract_network(network_to_otensors(tn), replace)
+            np.testing.assert_allclose(
+                total, np.atleast_1d(direct.data), rtol=1e-10, atol=1e-14)
+            result_queue.put("ok")
+    except Exception as e:  # pragma: no cover
+        if rank == 0:
+            result_queue.put(f"FAIL: {e!r}")
+        raise
+    finally:
+        dist_t.destroy_process_group()
+
+
+def test_sliced_allreduce_two_ranks_gloo():
+    """The multi-GPU bench's slicing scheme on CPU: ranks contract their
+    slice assignments, one all_reduce sums them; equals the direct
+    contraction."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_slice_worker, args=(r, 2, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    try:
+        verdict = q.get(timeout=180)
+    finally:
+        for p in procs:
+            p.join(timeout=60)
+            if p.is_alive():
+                p.terminate()
+    assert verdict == "ok", verdict

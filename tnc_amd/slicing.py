@@ -82,7 +82,11 @@ def find_slice_edges(tn: CompositeTensor, replace_toplevel,
             break  # nothing left to slice
         edges.append(max(score, key=score.get))
         new_peak, inters, _ = _walk_sizes(tn, replace_toplevel, skip=edges)
-        if new_peak >= peak:  # no progress — undo and stop
+        # chasing a memory target: stop when an edge makes no progress.
+        # target <= 1 means "give me max_edges edges for parallelism" —
+        # an edge in near-peak intermediates still cuts per-slice flops
+        # even when another same-size intermediate pins the peak.
+        if new_peak >= peak and target_peak_elems > 1:
             edges.pop()
             break
         peak = new_peak
