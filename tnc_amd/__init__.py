@@ -28,6 +28,8 @@ from .qasm import import_qasm
 from .partition import find_partitioning, partition_tensor_network
 from .repartition import (
     CommunicationScheme,
+    IntermediatePartitioningModel,
+    LeafPartitioningModel,
     NaivePartitioningModel,
     balance_partitions,
     compute_solution,
