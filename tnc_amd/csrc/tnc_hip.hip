@@ -2161,15 +2161,12 @@ extern "C" void* tn_net_result_dev(tn_net* net) {
 
 extern "C" int tn_net_pool_bytes(tn_net* net, u64* in_use, u64* cached) {
   if (!net) FAILV(TN_ERR_INVALID, "null net");
-  hipMemPool_t pool;
-  u64 used = 0, reserved = 0;
-  if (hipDeviceGetDefaultMemPool(&pool, net->device) == hipSuccess) {
-    (void)hipMemPoolGetAttribute(pool, hipMemPoolAttrUsedMemCurrent, &used);
-    (void)hipMemPoolGetAttribute(pool, hipMemPoolAttrReservedMemCurrent,
-                                 &reserved);
-  }
+  // arena telemetry (the stream-ordered pool is unused — see ws_alloc)
+  u64 used = 0;
+  for (const auto& b : net->arena.blocks)
+    if (!b.free) used += b.sz;
   if (in_use) *in_use = used;
-  if (cached) *cached = reserved;
+  if (cached) *cached = net->arena.size;
   return TN_OK;
 }
 
