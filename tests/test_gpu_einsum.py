@@ -245,3 +245,29 @@ def test_c64_network_engine():
 
     ref = contract_network(network_to_otensors(tn), replace)
     np.testing.assert_allclose(data, ref.data, rtol=1e-3, atol=1e-5)
+
+
+def test_fuzz_random_einsums():
+    """30 seeded random (labels, dims) configurations across the dispatch
+    space: shared/unshared legs in arbitrary interleavings, dims 1..8
+    (pow2 and not), rank 0..6 — each vs the oracle."""
+    rng = np.random.default_rng(123)
+    for case in range(30):
+        ra = int(rng.integers(0, 7))
+        rb = int(rng.integers(0, 7))
+        pool = list(range(14))  # ra + rb-only <= 12 < 14, so choices never run dry
+        a_labels = list(rng.choice(pool, size=ra, replace=False))
+        # share a random subset of A's labels, in shuffled positions
+        nshare = int(rng.integers(0, min(ra, rb) + 1)) if min(ra, rb) else 0
+        shared = list(rng.choice(a_labels, size=nshare, replace=False)) \
+            if nshare else []
+        b_only = [l for l in rng.choice(
+            [p for p in pool if p not in a_labels],
+            size=rb - nshare, replace=False)] if rb - nshare else []
+        b_labels = shared + b_only
+        rng.shuffle(b_labels)
+        dims = {l: int(rng.choice([1, 2, 3, 4, 8])) for l in pool}
+        a_shape = [dims[l] for l in a_labels]
+        b_shape = [dims[l] for l in b_labels]
+        run_case(a_labels, a_shape, b_labels, b_shape,
+                 seed=1000 + case, rtol=1e-10)
