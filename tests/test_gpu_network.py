@@ -250,3 +250,24 @@ def test_sliced_contraction_gpu():
     assert legs_s == legs_d == ref.legs
     np.testing.assert_allclose(sliced, direct, rtol=1e-12, atol=1e-13)
     np.testing.assert_allclose(sliced, ref.data, rtol=1e-10, atol=1e-12)
+
+
+def test_peps_gpu_vs_oracle():
+    """Data-filled PEPS sandwich (2x3, one PEPO layer) contracts on the GPU
+    to the oracle's scalar."""
+    from tnc_amd import Greedy
+    from tnc_amd.builders import peps
+    from tnc_amd.executor import contract_tensor_network_gpu
+    from tnc_amd.tensor import TensorData
+
+    rng = np.random.default_rng(5)
+    tn = peps(2, 3, 2, 2, 1)
+    for t in tn.tensors:
+        data = (rng.standard_normal(t.shape) +
+                1j * rng.standard_normal(t.shape))
+        t.set_tensor_data(TensorData(TensorData.MATRIX, matrix=data))
+    replace = Greedy().find_path(tn).replace_path()
+    legs, data = contract_tensor_network_gpu(tn, replace)
+    ref = contract_network(network_to_otensors(tn), replace)
+    assert legs == [] == ref.legs
+    np.testing.assert_allclose(data, ref.data, rtol=1e-10)

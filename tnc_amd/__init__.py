@@ -23,7 +23,7 @@ from .cost import (
 from .paths import Greedy, Optimal, PartitionSearch, RandomGreedy, BasicContractionPathResult
 from .circuit import Circuit, Permutor
 from .connectivity import ConnectivityLayout, connectivity_edges
-from .builders import random_circuit, sycamore_circuit
+from .builders import peps, random_circuit, sycamore_circuit
 from .qasm import import_qasm
 from .partition import find_partitioning, partition_tensor_network
 from .repartition import (

@@ -19,7 +19,7 @@ from .connectivity import (
     SYCAMORE_D,
     connectivity_edges,
 )
-from .tensor import CompositeTensor, TensorData
+from .tensor import CompositeTensor, LeafTensor, TensorData
 
 
 def _rng(seed_or_rng):
@@ -187,3 +187,63 @@ def random_circuit_with_set_observable(
             initial.extend([left, right])
     tn.push_tensors(initial)
     return tn
+
+
+def peps(length, depth, physical_dim, virtual_dim, layers):
+    """PEPS sandwich structure <bra| PEPO^layers |ket> on a length x depth
+    lattice (builders/peps.rs:446-467). Leaves carry legs/dims only (no
+    data), like the reference. Edge numbering per layer block of
+    total = p + vv + vh edges (p = L*D physical, vv = (L-1)*D row-internal
+    virtual, vh = (D-1)*L column virtual; peps.rs doc comment): site (i, j)
+    legs are [physical...] then row bonds (j-1, j) then column bonds
+    (i-1, i), absent neighbors skipped.
+
+    `layers` = 0 gives the inner product of two states."""
+    assert length > 1, "PEPS should have length greater than 1"
+    assert depth > 1, "PEPS should have depth greater than 1"
+    L, D = length, depth
+    p = L * D
+    vv = (L - 1) * D
+    vh = (D - 1) * L
+    total = p + vv + vh
+
+    def site_legs(i, j, phys, vbase):
+        """phys: list of physical legs; vbase: base of this layer's virtual
+        block (row bonds at vbase, column bonds at vbase + vv)."""
+        legs = list(phys)
+        dims = [physical_dim] * len(phys)
+        if j > 0:
+            legs.append(vbase + i * (L - 1) + j - 1)
+            dims.append(virtual_dim)
+        if j < L - 1:
+            legs.append(vbase + i * (L - 1) + j)
+            dims.append(virtual_dim)
+        if i > 0:
+            legs.append(vbase + vv + (i - 1) * L + j)
+            dims.append(virtual_dim)
+        if i < D - 1:
+            legs.append(vbase + vv + i * L + j)
+            dims.append(virtual_dim)
+        return LeafTensor(legs, dims)
+
+    tensors = []
+    # |ket> layer: physical legs [0, p), virtuals at [p, total)
+    for i in range(D):
+        for j in range(L):
+            tensors.append(site_legs(i, j, [i * L + j], p))
+    # PEPO layers k: physical prev = k*total + index, next = (k+1)*total +
+    # index, virtuals at (k+1)*total + p
+    for k in range(layers):
+        last, start = k * total, (k + 1) * total
+        for i in range(D):
+            for j in range(L):
+                idx = i * L + j
+                tensors.append(
+                    site_legs(i, j, [last + idx, start + idx], start + p))
+    # <bra| layer: physical = layers*total + index, virtuals directly at
+    # (layers+1)*total (the final block has no further physical edges)
+    last, start = layers * total, (layers + 1) * total
+    for i in range(D):
+        for j in range(L):
+            tensors.append(site_legs(i, j, [last + i * L + j], start))
+    return CompositeTensor(tensors)
