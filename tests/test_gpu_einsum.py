@@ -271,3 +271,34 @@ def test_fuzz_random_einsums():
         b_shape = [dims[l] for l in b_labels]
         run_case(a_labels, a_shape, b_labels, b_shape,
                  seed=1000 + case, rtol=1e-10)
+
+
+def test_fuzz_random_einsums_c64():
+    """12 seeded random configurations through the c64 path."""
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(77)
+    for case in range(12):
+        ra = int(rng.integers(1, 6))
+        rb = int(rng.integers(1, 6))
+        pool = list(range(12))
+        a_labels = list(rng.choice(pool, size=ra, replace=False))
+        nshare = int(rng.integers(0, min(ra, rb) + 1))
+        shared = list(rng.choice(a_labels, size=nshare, replace=False)) \
+            if nshare else []
+        b_only = [l for l in rng.choice(
+            [p for p in pool if p not in a_labels],
+            size=rb - nshare, replace=False)] if rb - nshare else []
+        b_labels = shared + b_only
+        rng.shuffle(b_labels)
+        dims = {l: int(rng.choice([1, 2, 4, 8])) for l in pool}
+        r2 = np.random.default_rng(2000 + case)
+        a = _rand([dims[l] for l in a_labels], r2).astype(np.complex64)
+        b = _rand([dims[l] for l in b_labels], r2).astype(np.complex64)
+        out_labels, _ = symmetric_difference(a_labels, list(a.shape),
+                                             b_labels, list(b.shape))
+        ref = oracle.contract_ndarrays(out_labels, a_labels,
+                                       a.astype(np.complex128), b_labels,
+                                       b.astype(np.complex128))
+        got = hiplib.einsum_c64(out_labels, a_labels, a, b_labels, b)
+        np.testing.assert_allclose(got, ref, rtol=2e-3, atol=1e-4)
