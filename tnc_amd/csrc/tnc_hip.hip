@@ -205,6 +205,70 @@ __global__ void k_einsum_anyk(const CT* __restrict__ A,
   }
 }
 
+// Tiled bit-permutation copy for pow2 full-span permutes (the TTGT packs
+// and unpack): dst[p] = src[spread(p)]. k_permute_ct reads src scattered
+// (one 8/16 B element per lane); here the 6 lowest-src-stride bits are the
+// lane index of a coalesced READ and the 6 lowest free dst bits the lane
+// index of a coalesced WRITE, staged through a 64x64 XOR-swizzled LDS tile
+// (64 KB c128 -> 2 blocks/CU).
+#define TN_PERM_AB 6
+#define TN_PERM_BB 6
+#define TN_PERM_MAXBITS 34
+
+struct PermPerm {
+  int rbits;
+  u64 restS[TN_PERM_MAXBITS];  // src / dst element strides of each rest bit
+  u64 restD[TN_PERM_MAXBITS];
+  u64 bS[TN_PERM_BB];  // src strides of the b-bits
+  u64 bD[TN_PERM_BB];  // dst strides of the b-bits (ascending)
+  u64 aD[TN_PERM_AB];  // dst strides of the a-bits (src strides are 1<<i)
+};
+
+template <typename CT>
+__global__ __launch_bounds__(512) void k_permute_tile(
+    const CT* __restrict__ src, CT* __restrict__ dst, PermPerm pp) {
+  __shared__ CT tile[64 * 64];
+  __shared__ u64 sboffS[64], sboffD[64], saoffD[64];
+  const int tid = threadIdx.x;
+  if (tid < 64) {
+    u64 os = 0, od = 0;
+    for (int i = 0; i < TN_PERM_BB; ++i)
+      if (tid >> i & 1) {
+        os += pp.bS[i];
+        od += pp.bD[i];
+      }
+    sboffS[tid] = os;
+    sboffD[tid] = od;
+  } else if (tid < 128) {
+    const int a = tid - 64;
+    u64 od = 0;
+    for (int i = 0; i < TN_PERM_AB; ++i)
+      if (a >> i & 1) od += pp.aD[i];
+    saoffD[a] = od;
+  }
+  u64 baseS = 0, baseD = 0;
+  {
+    unsigned r = blockIdx.x;
+    for (int i = 0; i < pp.rbits; ++i) {
+      if (r & 1) {
+        baseS += pp.restS[i];
+        baseD += pp.restD[i];
+      }
+      r >>= 1;
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < 64 * 64; e += 512) {
+    const int a = e & 63, b = e >> 6;  // lane ~ a: coalesced src reads
+    tile[b * 64 + (a ^ b)] = src[baseS + sboffS[b] + (u64)a];
+  }
+  __syncthreads();
+  for (int e = tid; e < 64 * 64; e += 512) {
+    const int b = e & 63, a = e >> 6;  // lane ~ b: coalesced dst writes
+    dst[baseD + saoffD[a] + sboffD[b]] = tile[b * 64 + (a ^ b)];
+  }
+}
+
 // fast path: both operands contiguous and identically ordered over the
 // contracted legs (the final amplitude dot of two same-legs tensors) —
 // pure streaming, no index gather.
@@ -1059,6 +1123,63 @@ static int grid_for(u64 nout, int block = 256) {
   return (int)blocks;
 }
 
+// Launch the tiled permute when the axis list describes a pow2 full-span
+// bit permutation (contiguous source, >= 2^20 elements); returns false to
+// fall back to the gather permute (strided views, non-pow2, tiny sizes).
+template <typename CT>
+static bool permute_tiled(const CT* src, CT* dst, u64 elems,
+                          const std::vector<AxisInfo>& ax,
+                          hipStream_t stream) {
+  if (elems < (1ull << 20) || elems > (1ull << TN_PERM_MAXBITS)) return false;
+  struct BitSD {
+    u64 ss, sd;
+  };
+  std::vector<BitSD> bits;
+  u64 dstride = 1;
+  for (int i = (int)ax.size() - 1; i >= 0; --i) {
+    const AxisInfo& a = ax[i];
+    if ((a.dim & (a.dim - 1)) || a.sa <= 0) return false;
+    for (u64 d = 1; d < a.dim; d <<= 1)
+      bits.push_back({(u64)a.sa * d, dstride * d});
+    dstride *= a.dim;
+  }
+  const int nb = (int)bits.size();
+  if (nb <= TN_PERM_AB + TN_PERM_BB || (1ull << nb) != elems) return false;
+  std::vector<int> byS(nb), byD(nb);
+  for (int i = 0; i < nb; ++i) byS[i] = byD[i] = i;
+  std::sort(byS.begin(), byS.end(),
+            [&](int x, int y) { return bits[x].ss < bits[y].ss; });
+  std::sort(byD.begin(), byD.end(),
+            [&](int x, int y) { return bits[x].sd < bits[y].sd; });
+  for (int i = 0; i < nb; ++i)
+    if (bits[byS[i]].ss != (1ull << i) || bits[byD[i]].sd != (1ull << i))
+      return false;  // not a contiguous span on both sides
+  PermPerm pp;
+  std::vector<char> used(nb, 0);
+  for (int i = 0; i < TN_PERM_AB; ++i) {
+    used[byS[i]] = 1;
+    pp.aD[i] = bits[byS[i]].sd;
+  }
+  int nbb = 0, nr = 0;
+  for (int i = 0; i < nb && nbb < TN_PERM_BB; ++i) {
+    const int bi = byD[i];
+    if (used[bi]) continue;
+    pp.bS[nbb] = bits[bi].ss;
+    pp.bD[nbb] = bits[bi].sd;
+    used[bi] = 1;
+    ++nbb;
+  }
+  for (int i = 0; i < nb; ++i) {
+    if (used[byS[i]]) continue;
+    pp.restS[nr] = bits[byS[i]].ss;
+    pp.restD[nr] = bits[byS[i]].sd;
+    ++nr;
+  }
+  pp.rbits = nr;
+  k_permute_tile<<<dim3(1u << nr), 512, 0, stream>>>(src, dst, pp);
+  return true;
+}
+
 // core einsum over device buffers; out is contiguous row-major in out order.
 template <typename CT>
 static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
@@ -1341,12 +1462,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       if (rc_ == TN_ERR_OOM && gather_ok) return run_gather();
       if (rc_) return rc_;
     }
-    int blocks = grid_for(elems);
-    if (map.pow2)
-      k_permute_ct<true><<<blocks, 256, 0, stream>>>(Adata, packA, elems, map);
-    else
-      k_permute_ct<false><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
-                                                      map);
+    if (!permute_tiled(Adata, packA, elems, ax, stream)) {
+      int blocks = grid_for(elems);
+      if (map.pow2)
+        k_permute_ct<true><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
+                                                       map);
+      else
+        k_permute_ct<false><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
+                                                        map);
+    }
     Ag = packA;
   }
   if (!is_ready(B, b_axes)) {
@@ -1363,12 +1487,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       }
       if (rc_) return rc_;
     }
-    int blocks = grid_for(elems);
-    if (map.pow2)
-      k_permute_ct<true><<<blocks, 256, 0, stream>>>(Bdata, packB, elems, map);
-    else
-      k_permute_ct<false><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
-                                                      map);
+    if (!permute_tiled(Bdata, packB, elems, ax, stream)) {
+      int blocks = grid_for(elems);
+      if (map.pow2)
+        k_permute_ct<true><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
+                                                       map);
+      else
+        k_permute_ct<false><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
+                                                        map);
+    }
     Bg = packB;
   }
 
@@ -1477,15 +1604,17 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     std::vector<AxisInfo> ax;
     for (int i = 0; i < out_nd; ++i)
       ax.push_back({out_shape[i], tmp_stride[i], 0});
-    GatherMap map;
-    if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
-    int blocks = grid_for(nout);
-    if (map.pow2)
-      k_permute_ct<true><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
-                                                     nout, map);
-    else
-      k_permute_ct<false><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
-                                                      nout, map);
+    if (!permute_tiled((const CT*)tmpC, out, nout, ax, stream)) {
+      GatherMap map;
+      if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
+      int blocks = grid_for(nout);
+      if (map.pow2)
+        k_permute_ct<true><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
+                                                       nout, map);
+      else
+        k_permute_ct<false><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
+                                                        nout, map);
+    }
     HIP_CHECK(hipGetLastError());
   }
   ws_free(ws, packA);
