@@ -19,6 +19,7 @@ from .cost import (
     contract_size_tensors,
     contract_path_cost,
     communication_path_cost,
+    communication_path_op_costs,
 )
 from .paths import Greedy, Optimal, PartitionSearch, RandomGreedy, BasicContractionPathResult
 from .circuit import Circuit, Permutor

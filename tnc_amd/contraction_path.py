@@ -33,6 +33,15 @@ class ContractionPath:
     def is_simple(self):
         return not self.nested
 
+    def is_empty(self):
+        return not self.toplevel and not self.nested
+
+    def into_simple(self):
+        """The toplevel replace path; asserts no nested parts
+        (contractionpath.rs into_simple semantics)."""
+        assert self.is_simple(), "path has nested parts"
+        return self.toplevel
+
     def __eq__(self, other):
         return (
             isinstance(other, ContractionPath)

@@ -96,3 +96,14 @@ def compute_memory_requirements(inputs, p: ContractionPath, memory_estimator):
     """contraction_cost.rs:254-264."""
     _, mem = _contract_path_custom_cost(inputs, p, lambda a, b: 0.0, memory_estimator)
     return mem
+
+
+def communication_path_op_costs(inputs, contract_path, only_count_ops,
+                                tensor_cost=None):
+    """((parallel, serial), mem): critical-path and sum time complexity plus
+    space complexity (contraction_cost.rs:156-167)."""
+    parallel, _ = communication_path_cost(inputs, contract_path,
+                                          only_count_ops, True, tensor_cost)
+    serial, mem = communication_path_cost(inputs, contract_path,
+                                          only_count_ops, False, tensor_cost)
+    return (parallel, serial), mem
