@@ -141,3 +141,18 @@ def partition_tensor_network(tn: CompositeTensor, partitioning) -> CompositeTens
     for p, t in zip(partitioning, tn.tensors):
         parts[index[p]].push_tensor(t)
     return CompositeTensor(parts)
+
+
+def communication_partitioning(tensors, k: int, imbalance=0.03,
+                               strategy=PartitioningStrategy.MIN_CUT,
+                               minimize=True, seed=0) -> list:
+    """k-way partition of weighted (cost, LeafTensor) tuples for
+    communication-scheme search (partitioning.rs:100-160): min-cut groups
+    tensors sharing heavy edges together, max-cut (`minimize=False`)
+    spreads them apart. Edge weights are log2(dim) like the reference's
+    scaled-log KaHyPar weights; max-cut negates them."""
+    assert k > 1, "Partitioning only valid for more than one process"
+    tn = CompositeTensor([t for _, t in tensors])
+    part = find_partitioning(tn, k, strategy=strategy, minimize=minimize,
+                             seed=seed, imbalance=imbalance)
+    return part

@@ -25,3 +25,8 @@ def random_sparse_tensor_data(dims, sparsity=None, rng=None) -> TensorData:
         data[loc] = complex(rng.random(), rng.random())
         nnz += 1
     return TensorData(TensorData.MATRIX, matrix=data)
+
+
+def random_sparse_tensor_data_with_rng(dims, sparsity, rng) -> TensorData:
+    """tensorgeneration.rs:18-41 (explicit-rng variant)."""
+    return random_sparse_tensor_data(dims, sparsity, rng)
