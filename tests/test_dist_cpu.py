@@ -274,14 +274,11 @@ def _slice_worker(rank, world, result_queue):
         dist_t.destroy_process_group()
 
 
-def test_sliced_allreduce_two_ranks_gloo():
-    """The multi-GPU bench's slicing scheme on CPU: ranks contract their
-    slice assignments, one all_reduce sums them; equals the direct
-    contraction."""
+def _run_sliced(world):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_slice_worker, args=(r, 2, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_slice_worker, args=(r, world, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     try:
@@ -292,3 +289,15 @@ def test_sliced_allreduce_two_ranks_gloo():
             if p.is_alive():
                 p.terminate()
     assert verdict == "ok", verdict
+
+
+def test_sliced_allreduce_two_ranks_gloo():
+    """The multi-GPU bench's slicing scheme on CPU: ranks contract their
+    slice assignments, one all_reduce sums them; equals the direct
+    contraction."""
+    _run_sliced(2)
+
+
+def test_sliced_allreduce_four_ranks_gloo():
+    """Four ranks, two sliced edges (the N=4 shape of the scale run)."""
+    _run_sliced(4)
