@@ -2222,9 +2222,11 @@ extern "C" int tn_net_contract(tn_net* net, const u64* pairs, size_t nsteps,
                                double* elapsed_ms) {
   if (!net) FAILV(TN_ERR_INVALID, "null net");
   if (net->graph_state == 2) {
-    if (graph_pairs_match(net, pairs, nsteps))
+    // a failed walk since capture clears final_t, whose address the replay
+    // relies on — in that case re-run (and re-capture) normally
+    if (graph_pairs_match(net, pairs, nsteps) && net->final_t.data)
       return contract_graph_replay(net, elapsed_ms);
-    net->invalidate_graph();  // different path: baked pointers are stale
+    net->invalidate_graph();  // different path / lost state: re-arm below
   }
   if (net->graph_state == 1 && net->arena.base &&
       graph_pairs_match(net, pairs, nsteps)) {
