@@ -1138,6 +1138,7 @@ static bool permute_tiled(const CT* src, CT* dst, u64 elems,
   u64 dstride = 1;
   for (int i = (int)ax.size() - 1; i >= 0; --i) {
     const AxisInfo& a = ax[i];
+    if (a.dim == 1) continue;  // contributes no bits; stride irrelevant
     if ((a.dim & (a.dim - 1)) || a.sa <= 0) return false;
     for (u64 d = 1; d < a.dim; d <<= 1)
       bits.push_back({(u64)a.sa * d, dstride * d});
@@ -1288,6 +1289,7 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       std::vector<BitSB> bits;
       bool ok = true;
       for (auto& ax : kax) {
+        if (ax.dim == 1) continue;  // contributes no bits
         if ((ax.dim & (ax.dim - 1)) || ax.sa <= 0 || ax.sb <= 0) {
           ok = false;
           break;
