@@ -14,7 +14,7 @@ import numpy as np
 from . import hiplib
 from .contraction_path import ContractionPath, flatten_network
 from .cost import contract_cost_tensors
-from .tensor import CompositeTensor, LeafTensor, TensorData
+from .tensor import CompositeTensor, LeafTensor
 
 
 class StepInfo:
