@@ -120,7 +120,8 @@ class ContractionEngine:
             raise RuntimeError(f"tn_net_create failed: {hiplib.last_error()}")
         reserve = arena_bytes(leaves, steps, self.infos, self.esize)
         # arena is an optimization: if the reservation fails (tiny GPUs,
-        # fragmented memory), the executor falls back to hipMallocAsync
+        # fragmented memory), every block falls back to plain hipMalloc
+        # (the stream-ordered pool is avoided — see ws_alloc in tnc_hip.hip)
         if reserve > 64 * 1024 * 1024:
             rc = L.tn_net_reserve(self.net, reserve)
             if rc != 0:
@@ -128,7 +129,8 @@ class ContractionEngine:
 
                 warnings.warn(
                     f"arena reservation of {reserve} bytes failed "
-                    f"({hiplib.last_error()}); falling back to async allocs"
+                    f"({hiplib.last_error()}); falling back to per-block "
+                    "hipMalloc"
                 )
         for t in leaves:
             data = np.ascontiguousarray(t.tensordata.into_data(), dtype=self.npdtype)
