@@ -31,9 +31,10 @@ def main():
         # best-of search (the config-3 path is "the best path
         # found" — RandomGreedy(many) + the partition-guided quality tier
         # standing in for cotengra HyperOptimizer)
-        rg = PartitionSearch(ks=(2, 3, 4, 6, 8, 12, 16), seeds=(0, 1, 2, 3),
-                             trials=cfg["trials"],
-                             size_cap=cfg.get("size_cap")).find_path(tn)
+        finder_args = dict(ks=(2, 3, 4, 6, 8, 12, 16), seeds=(0, 1, 2, 3),
+                           trials=cfg["trials"],
+                           size_cap=cfg.get("size_cap"))
+        rg = PartitionSearch(**finder_args).find_path(tn)
         replace = rg.replace_path()
         assert not replace.nested
         flops = metric_flops(tn, replace.toplevel)
@@ -43,6 +44,16 @@ def main():
             "op_cost": rg.flops,
             "peak_size_elems": rg.size,
             "metric_flops": flops,
+            # provenance of the frozen path (surfaced by bench.py's
+            # config.path_provenance)
+            "path_finder": {
+                "finder": "PartitionSearch",
+                "ks": list(finder_args["ks"]),
+                "seeds": list(finder_args["seeds"]),
+                "trials": finder_args["trials"],
+                "size_cap": finder_args["size_cap"],
+                "greedy_seed": 42,
+            },
         }
         save_network(tn, fixture_path(name), replace.toplevel, meta)
         print(

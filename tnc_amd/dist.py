@@ -248,26 +248,27 @@ def make_tree_plan(tn: CompositeTensor, replace_toplevel, nranks: int) -> DistPl
 
 
 def make_plan(tn: CompositeTensor, nranks: int, trials: int = 16,
-              size_cap=None, seed: int = 0, sa_seconds: float = 0.0) -> DistPlan:
+              size_cap=None, seed: int = 0, sa_rounds: int = 0) -> DistPlan:
     """Partition + per-partition paths + fan-in path, all deterministic.
 
-    sa_seconds > 0 refines the initial min-cut partitioning with the
+    sa_rounds > 0 refines the initial min-cut partitioning with the
     reference's simulated-annealing repartitioner (simulated_annealing.rs
-    semantics; see repartition.py) before path finding. Note: SA wall-time
-    budgets make the plan timing-dependent — every rank must either use
-    sa_seconds=0 or receive the same refined partitioning (bench.py runs SA
-    on rank 0's inputs deterministically by seeding and step count)."""
+    semantics; see repartition.py) before path finding, running exactly
+    sa_rounds SA rounds so that independent ranks seeding the same rng
+    derive the SAME plan (a wall-clock budget would make the plan
+    timing-dependent and desynchronize rank wire shapes)."""
     if nranks == 1:
         raise ValueError("use the single-GPU engine for one rank")
     partitioning = find_partitioning(tn, nranks, seed=seed)
-    if sa_seconds > 0:
+    if sa_rounds > 0:
         import numpy as np
 
         from .repartition import NaivePartitioningModel, balance_partitions
 
         model = NaivePartitioningModel(tn, nranks)
         partitioning, _ = balance_partitions(
-            model, partitioning, np.random.default_rng(seed), sa_seconds
+            model, partitioning, np.random.default_rng(seed),
+            n_rounds=sa_rounds
         )
     ptn = partition_tensor_network(tn, partitioning)
     result = RandomGreedy(trials, size_cap=size_cap).find_path(ptn)

@@ -82,7 +82,7 @@ def _greedy_ssa(leaves: List[LeafTensor], costmod=1.0, temperature=0.0, rng=None
     def push_candidate(i, j):
         nonlocal counter
         tij = legs[i] ^ legs[j]
-        score = tij.size() / costmod - (sizes[i] + sizes[j]) * costmod
+        score = tij.size() - costmod * (sizes[i] + sizes[j])
         heapq.heappush(heap, (perturb(score), counter, i, j, tij))
         counter += 1
 
@@ -285,15 +285,19 @@ class Optimal(_CotengrustLike):
             ssa = []
             next_id = n
             ids = []
+            id_view = {}
             for comp in comps:
                 cid, next_id = self._emit(comp, best, ssa, n, next_id)
+                id_view[cid] = views[comp]
                 ids.append((views[comp].size(), -cid, cid))
             heapq.heapify(ids)
             while len(ids) > 1:
                 _, _, a = heapq.heappop(ids)
                 _, _, b = heapq.heappop(ids)
                 ssa.append((a, b))
-                heapq.heappush(ids, (1.0, -next_id, next_id))
+                tab = id_view[a] ^ id_view[b]
+                id_view[next_id] = tab
+                heapq.heappush(ids, (tab.size(), -next_id, next_id))
                 next_id += 1
             return ssa
         ssa = []

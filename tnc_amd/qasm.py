@@ -229,7 +229,16 @@ def import_qasm(code: str) -> Circuit:
         return base + index
 
     def apply_gate(name, angle_vals, qubit_ids, adjoint_ctx=False):
-        # user-defined gates inline recursively (reference: gate inlining)
+        # registry builtins win over same-named user declarations (the
+        # reference's gate inliner keeps registry gates regardless of
+        # declarations: ast.rs is_builtin precedence); only unknown names
+        # inline recursively from their declaration body
+        if name in _STD_GATES:
+            gname, angles, adjoint = _STD_GATES[name](angle_vals)
+            qubits = [circuit.qubit(q) for q in qubit_ids]
+            circuit.append_gate(TensorData.from_gate(gname, angles, adjoint),
+                                qubits)
+            return
         if name in user_gates:
             gd = user_gates[name]
             assert len(angle_vals) == len(gd.params)
@@ -241,11 +250,7 @@ def import_qasm(code: str) -> Circuit:
                 bqubits = [qmap[q] for q in bqargs]
                 apply_gate(bname, bangles, bqubits)
             return
-        if name not in _STD_GATES:
-            raise ValueError(f"unknown gate '{name}'")
-        gname, angles, adjoint = _STD_GATES[name](angle_vals)
-        qubits = [circuit.qubit(q) for q in qubit_ids]
-        circuit.append_gate(TensorData.from_gate(gname, angles, adjoint), qubits)
+        raise ValueError(f"unknown gate '{name}'")
 
     while i < len(toks):
         kind, val = toks[i]

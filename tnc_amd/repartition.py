@@ -302,11 +302,21 @@ class IntermediatePartitioningModel:
                                       self.memory_limit, rng)
 
 
-def balance_partitions(model, initial_solution, rng, max_time_s,
+def balance_partitions(model, initial_solution, rng, max_time_s=None,
                        n_trials=8, n_steps=40, restart_iter=50,
-                       initial_temperature=2.0, final_temperature=0.05):
+                       initial_temperature=2.0, final_temperature=0.05,
+                       n_rounds=None):
     """simulated_annealing.rs:85-166, 576-595: SA with log2-ratio acceptance
-    and time-budgeted log-interpolated temperature."""
+    and a log-interpolated temperature schedule.
+
+    Schedule: either wall-clock budgeted over `max_time_s` (the reference's
+    mode, simulated_annealing.rs:586-593 — timing-dependent, so the result is
+    NOT reproducible across machines) or, with `n_rounds` set, exactly
+    `n_rounds` outer rounds with the temperature interpolated over round
+    index — fully deterministic for a given rng seed, which distributed
+    plan derivation requires (every rank must compute the same plan)."""
+    assert (max_time_s is not None) != (n_rounds is not None), \
+        "pass exactly one of max_time_s / n_rounds"
     rng = np.random.default_rng(rng) if not isinstance(rng, np.random.Generator) else rng
     current = list(initial_solution)
     current_score = model.evaluate(current, rng)
@@ -316,8 +326,9 @@ def balance_partitions(model, initial_solution, rng, max_time_s,
     steps_per_chain = max(1, -(-n_steps // n_trials))
     log_start = math.log2(initial_temperature)
     log_end = math.log2(final_temperature)
-    t_end = time.monotonic() + max_time_s
+    t_end = time.monotonic() + max_time_s if max_time_s is not None else None
     temperature = initial_temperature
+    rounds = 0
     while True:
         chain_results = []
         for c in range(n_trials):
@@ -344,9 +355,15 @@ def balance_partitions(model, initial_solution, rng, max_time_s,
         last_improvement += 1
         if last_improvement == restart_iter:
             current, current_score = list(best), best_score
-        now = time.monotonic()
-        if now > t_end:
-            break
-        progress = 1.0 - (t_end - now) / max_time_s
+        rounds += 1
+        if n_rounds is not None:
+            if rounds >= n_rounds:
+                break
+            progress = rounds / n_rounds
+        else:
+            now = time.monotonic()
+            if now > t_end:
+                break
+            progress = 1.0 - (t_end - now) / max_time_s
         temperature = 2.0 ** (log_start + (log_end - log_start) * progress)
     return best, best_score
