@@ -302,3 +302,114 @@ def test_fuzz_random_einsums_c64():
                                        b.astype(np.complex128))
         got = hiplib.einsum_c64(out_labels, a_labels, a, b_labels, b)
         np.testing.assert_allclose(got, ref, rtol=2e-3, atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# Bit-exact index permutations (north star: "bit-exact for index
+# permutations"). Multiplication by exactly 1.0+0.0j and accumulation of
+# exact zeros are lossless in IEEE754, so any einsum that is semantically a
+# permutation must reproduce the oracle's transpose EXACTLY — through the
+# gather kernels (scalar B) and through the full TTGT pack/GEMM/unpack
+# machinery (identity B), pow2 and non-pow2 dims.
+# ---------------------------------------------------------------------------
+
+
+def _exact_permute_scalar_b(a_labels, a_shape, out_labels, seed, c64=False):
+    """A x scalar-1 with a caller-chosen out order == np.transpose, exactly."""
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(seed)
+    a = _rand(a_shape, rng)
+    if c64:
+        a = a.astype(np.complex64)
+    one = np.ones((), dtype=a.dtype)
+    fn = hiplib.einsum_c64 if c64 else hiplib.einsum_c128
+    got = fn(out_labels, a_labels, a, [], one)
+    perm = [a_labels.index(l) for l in out_labels]
+    expect = np.transpose(a, perm)
+    np.testing.assert_array_equal(got, expect)
+
+
+def test_permute_scalar_b_exact_nonpow2():
+    _exact_permute_scalar_b([0, 1, 2, 3], [3, 5, 7, 2], [2, 0, 3, 1], seed=11)
+
+
+def test_permute_scalar_b_exact_pow2():
+    _exact_permute_scalar_b(list(range(10)), [2] * 10,
+                            [9, 0, 7, 1, 5, 2, 6, 3, 8, 4], seed=12)
+
+
+def test_permute_scalar_b_exact_c64():
+    _exact_permute_scalar_b([0, 1, 2], [6, 10, 9], [2, 0, 1], seed=13,
+                            c64=True)
+
+
+def _exact_permute_identity_b(m_dims, k_dims, seed, c64=False):
+    """A contracted with an identity over its K legs is a pure permutation:
+    the TTGT path (pack permutes k_permute_ct/k_permute_tile, MFMA GEMM,
+    split-K, unpack) must reproduce A's transpose bit-for-bit."""
+    from tnc_amd import hiplib
+
+    rng = np.random.default_rng(seed)
+    nm, nk = len(m_dims), len(k_dims)
+    m_labels = list(range(nm))
+    k_labels = [100 + i for i in range(nk)]
+    n_labels = [200 + i for i in range(nk)]
+    # interleave K legs into A's storage order to force a pack permute
+    a_labels, a_dims = [], []
+    for i in range(max(nm, nk)):
+        if i < nk:
+            a_labels.append(k_labels[i])
+            a_dims.append(k_dims[i])
+        if i < nm:
+            a_labels.append(m_labels[i])
+            a_dims.append(m_dims[i])
+    a = _rand(a_dims, rng)
+    K = int(np.prod(k_dims))
+    b = np.eye(K, dtype=np.complex128).reshape(tuple(k_dims) + tuple(k_dims))
+    b_labels = k_labels + n_labels
+    if c64:
+        a = a.astype(np.complex64)
+        b = b.astype(np.complex64)
+    # interleaved out order to force an unpack as well
+    out_labels = []
+    for i in range(max(nm, nk)):
+        if i < nk:
+            out_labels.append(n_labels[nk - 1 - i])
+        if i < nm:
+            out_labels.append(m_labels[i])
+    fn = hiplib.einsum_c64 if c64 else hiplib.einsum_c128
+    got = fn(out_labels, a_labels, a, b_labels, b)
+    # expected: A with K legs relabeled to N, transposed to the out order
+    relabel = dict(zip(k_labels, n_labels))
+    a_as_out = [relabel.get(l, l) for l in a_labels]
+    perm = [a_as_out.index(l) for l in out_labels]
+    expect = np.transpose(a, perm)
+    np.testing.assert_array_equal(got, expect)
+
+
+def test_permute_identity_b_gemm_exact_nonpow2():
+    # m = 296, n = k = 72: GEMM-worthy (k>=16, m>=128, n>=64), ragged dims
+    _exact_permute_identity_b([37, 8], [9, 8], seed=21)
+
+
+def test_permute_identity_b_gemm_exact_pow2():
+    # all dims 2, m = 256, n = k = 256: the bit-permutation tiled path
+    _exact_permute_identity_b([2] * 8, [2] * 8, seed=22)
+
+
+def test_permute_identity_b_gemm_exact_c64():
+    _exact_permute_identity_b([37, 8], [9, 8], seed=23, c64=True)
+
+
+def test_final_permutor_exact():
+    """The final Permutor (circuit_builder.rs:77-122) is an exact transpose."""
+    from tnc_amd import Permutor
+
+    rng = np.random.default_rng(31)
+    data = _rand([2, 3, 4, 5], rng)
+    legs = [7, 3, 9, 1]
+    p = Permutor([1, 9, 3, 7])
+    new_legs, new_dims, out = p.apply(legs, list(data.shape), data)
+    assert new_legs == [1, 9, 3, 7]
+    np.testing.assert_array_equal(out, np.transpose(data, [3, 2, 1, 0]))
