@@ -150,3 +150,48 @@ def test_bipartition_schemes():
             assert x not in consumed and y not in consumed or True
             consumed.add(y)
         assert len(consumed) == len(children) - 1
+
+
+def test_balance_partitions_workers_bit_identical():
+    """Per-chain seeds are drawn from the master rng before the chains run,
+    so the SA result is identical for any worker count (the reference's
+    48-thread rayon pool, simulated_annealing.rs:35-36, parallelized here
+    with a fork process pool)."""
+    import numpy as np
+
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.partition import find_partitioning
+    from tnc_amd.repartition import NaivePartitioningModel, balance_partitions
+
+    tn = random_circuit(10, 6, 0.5, 0.5, 3, ConnectivityLayout.EAGLE)
+    init = find_partitioning(tn, 3, seed=0)
+    model = NaivePartitioningModel(tn, 3)
+    serial, s_score = balance_partitions(
+        model, init, np.random.default_rng(7), n_rounds=6, n_trials=4,
+        workers=1)
+    parallel, p_score = balance_partitions(
+        model, init, np.random.default_rng(7), n_rounds=6, n_trials=4,
+        workers=2)
+    assert s_score == p_score
+    assert serial == parallel
+
+
+def test_balance_partitions_fixed_rounds_deterministic():
+    """n_rounds mode is reproducible run-to-run (required for distributed
+    plan derivation: every rank must compute the same refined plan)."""
+    import numpy as np
+
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.partition import find_partitioning
+    from tnc_amd.repartition import NaivePartitioningModel, balance_partitions
+
+    tn = random_circuit(10, 6, 0.5, 0.5, 4, ConnectivityLayout.EAGLE)
+    init = find_partitioning(tn, 2, seed=0)
+    model = NaivePartitioningModel(tn, 2)
+    a = balance_partitions(model, init, np.random.default_rng(9), n_rounds=5,
+                           n_trials=3)
+    b = balance_partitions(model, init, np.random.default_rng(9), n_rounds=5,
+                           n_trials=3)
+    assert a == b

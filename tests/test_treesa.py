@@ -1,0 +1,66 @@
+"""NNI tree-annealing refinement tier (tnc_amd/treesa.py): the
+HyperOptimizer-substitute quality pass on top of PartitionSearch
+(paths/hyperoptimization.rs:1-60 interface)."""
+
+import numpy as np
+
+from oracle import contract_network
+from oracle.adapters import network_to_otensors
+from tnc_amd import Greedy
+from tnc_amd.builders import random_circuit
+from tnc_amd.connectivity import ConnectivityLayout
+from tnc_amd.cost import contract_path_cost
+from tnc_amd.fixtures import load_fixture
+from tnc_amd.contraction_path import ContractionPath, validate_path
+from tnc_amd.tensor import LeafTensor
+from tnc_amd.treesa import TreeSA, refine_replace_path
+
+
+def test_refined_path_same_value():
+    """NNI moves only reshape the tree: the refined path contracts to the
+    same value as the input path."""
+    tn = random_circuit(12, 8, 0.5, 0.5, 7, ConnectivityLayout.EAGLE)
+    g = Greedy().find_path(tn)
+    ref = contract_network(network_to_otensors(tn), g.replace_path())
+    leaves = [LeafTensor(t.legs, t.bond_dims) for t in tn.tensors]
+    top, op, peak = refine_replace_path(
+        leaves, list(g.replace_path().toplevel), moves=20_000, seed=1)
+    out = contract_network(network_to_otensors(tn), top)
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)
+    # the incremental cost bookkeeping agrees with the standalone model
+    op2, _ = contract_path_cost(tn.tensors, ContractionPath.simple(top), True)
+    assert abs(op - op2) <= 1e-6 * op2
+
+
+def test_refines_frozen_rqc36():
+    """On the frozen headline fixture the refinement finds a strictly
+    cheaper tree under the same size cap (the recorded r02 sweep reached
+    -19% with chained restarts)."""
+    tn, rp, meta = load_fixture("rqc36")
+    leaves = [LeafTensor(t.legs, t.bond_dims) for t in tn.tensors]
+    top, op, peak = refine_replace_path(leaves, rp, moves=50_000, seed=0,
+                                        size_cap=6.0e9)
+    assert op < meta["op_cost"]
+    assert peak <= 6.0e9
+    validate_path(ContractionPath.simple(top))
+
+
+def test_never_worse_than_input():
+    tn = random_circuit(10, 6, 0.5, 0.5, 11, ConnectivityLayout.EAGLE)
+    g = Greedy().find_path(tn)
+    leaves = [LeafTensor(t.legs, t.bond_dims) for t in tn.tensors]
+    top, op, _ = refine_replace_path(
+        leaves, list(g.replace_path().toplevel), moves=2_000, seed=3)
+    assert op <= g.flops + 1e-9 * g.flops
+
+
+def test_treesa_finder_end_to_end():
+    tn = random_circuit(12, 8, 0.5, 0.6, 5, ConnectivityLayout.EAGLE)
+    base = Greedy()
+    res = TreeSA(base=base, moves=10_000, seed=2).find_path(tn)
+    validate_path(res.replace_path())
+    assert res.flops <= base.find_path(tn).flops
+    ref = contract_network(network_to_otensors(tn),
+                           base.find_path(tn).replace_path())
+    out = contract_network(network_to_otensors(tn), res.replace_path())
+    np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)

@@ -302,10 +302,42 @@ class IntermediatePartitioningModel:
                                       self.memory_limit, rng)
 
 
+def _run_chain(model, trial, trial_score, steps, temperature, rng):
+    """One SA candidate chain (the body each rayon thread runs in the
+    reference, simulated_annealing.rs:112-134)."""
+    for _ in range(steps):
+        cand = model.generate_trial_solution(trial, rng)
+        score = model.evaluate(cand, rng)
+        if (not math.isfinite(score) or not math.isfinite(trial_score)
+                or score <= 0 or trial_score <= 0):
+            accept = score <= trial_score
+        else:
+            diff = math.log2(score / trial_score)
+            accept = math.exp(min(50.0, -diff / temperature)) >= rng.random()
+        if accept:
+            trial = cand
+            trial_score = score
+    return trial_score, trial
+
+
+_CHAIN_MODEL = None
+
+
+def _chain_worker_init(model):
+    global _CHAIN_MODEL
+    _CHAIN_MODEL = model
+
+
+def _chain_task(args):
+    current, current_score, steps, temperature, seed = args
+    return _run_chain(_CHAIN_MODEL, list(current), current_score, steps,
+                      temperature, np.random.default_rng(seed))
+
+
 def balance_partitions(model, initial_solution, rng, max_time_s=None,
                        n_trials=8, n_steps=40, restart_iter=50,
                        initial_temperature=2.0, final_temperature=0.05,
-                       n_rounds=None):
+                       n_rounds=None, workers=1):
     """simulated_annealing.rs:85-166, 576-595: SA with log2-ratio acceptance
     and a log-interpolated temperature schedule.
 
@@ -314,7 +346,13 @@ def balance_partitions(model, initial_solution, rng, max_time_s=None,
     NOT reproducible across machines) or, with `n_rounds` set, exactly
     `n_rounds` outer rounds with the temperature interpolated over round
     index — fully deterministic for a given rng seed, which distributed
-    plan derivation requires (every rank must compute the same plan)."""
+    plan derivation requires (every rank must compute the same plan).
+
+    workers > 1 evaluates the candidate chains of each round in a fork
+    process pool (the reference's 48 rayon threads,
+    simulated_annealing.rs:35-36). Per-chain rng seeds are drawn from the
+    master rng BEFORE the chains run, so the result is bit-identical for
+    any worker count (including 1)."""
     assert (max_time_s is not None) != (n_rounds is not None), \
         "pass exactly one of max_time_s / n_rounds"
     rng = np.random.default_rng(rng) if not isinstance(rng, np.random.Generator) else rng
@@ -329,41 +367,48 @@ def balance_partitions(model, initial_solution, rng, max_time_s=None,
     t_end = time.monotonic() + max_time_s if max_time_s is not None else None
     temperature = initial_temperature
     rounds = 0
-    while True:
-        chain_results = []
-        for c in range(n_trials):
-            trial = list(current)
-            trial_score = current_score
-            for _ in range(steps_per_chain):
-                cand = model.generate_trial_solution(trial, rng)
-                score = model.evaluate(cand, rng)
-                if (not math.isfinite(score) or not math.isfinite(trial_score)
-                        or score <= 0 or trial_score <= 0):
-                    accept = score <= trial_score
-                else:
-                    diff = math.log2(score / trial_score)
-                    accept = math.exp(min(50.0, -diff / temperature)) >= rng.random()
-                if accept:
-                    trial = cand
-                    trial_score = score
-            chain_results.append((trial_score, c, trial))
-        trial_score, _, trial = min(chain_results, key=lambda x: (x[0], x[1]))
-        current, current_score = trial, trial_score
-        if current_score < best_score:
-            best, best_score = list(current), current_score
-            last_improvement = 0
-        last_improvement += 1
-        if last_improvement == restart_iter:
-            current, current_score = list(best), best_score
-        rounds += 1
-        if n_rounds is not None:
-            if rounds >= n_rounds:
-                break
-            progress = rounds / n_rounds
-        else:
-            now = time.monotonic()
-            if now > t_end:
-                break
-            progress = 1.0 - (t_end - now) / max_time_s
-        temperature = 2.0 ** (log_start + (log_end - log_start) * progress)
+    pool = None
+    if workers > 1:
+        import multiprocessing
+
+        ctx = multiprocessing.get_context("fork")
+        pool = ctx.Pool(workers, initializer=_chain_worker_init,
+                        initargs=(model,))
+    try:
+        while True:
+            seeds = [int(s) for s in rng.integers(0, 2 ** 63, size=n_trials)]
+            tasks = [(current, current_score, steps_per_chain, temperature, s)
+                     for s in seeds]
+            if pool is not None:
+                outs = pool.map(_chain_task, tasks)
+            else:
+                outs = [_run_chain(model, list(current), current_score,
+                                   steps_per_chain, temperature,
+                                   np.random.default_rng(s)) for s in seeds]
+            chain_results = [(score, c, trial)
+                             for c, (score, trial) in enumerate(outs)]
+            trial_score, _, trial = min(chain_results,
+                                        key=lambda x: (x[0], x[1]))
+            current, current_score = trial, trial_score
+            if current_score < best_score:
+                best, best_score = list(current), current_score
+                last_improvement = 0
+            last_improvement += 1
+            if last_improvement == restart_iter:
+                current, current_score = list(best), best_score
+            rounds += 1
+            if n_rounds is not None:
+                if rounds >= n_rounds:
+                    break
+                progress = rounds / n_rounds
+            else:
+                now = time.monotonic()
+                if now > t_end:
+                    break
+                progress = 1.0 - (t_end - now) / max_time_s
+            temperature = 2.0 ** (log_start + (log_end - log_start) * progress)
+    finally:
+        if pool is not None:
+            pool.close()
+            pool.join()
     return best, best_score
