@@ -64,3 +64,38 @@ def test_treesa_finder_end_to_end():
                            base.find_path(tn).replace_path())
     out = contract_network(network_to_otensors(tn), res.replace_path())
     np.testing.assert_allclose(out.data, ref.data, rtol=1e-10, atol=1e-12)
+
+
+def test_reduce_peak_small_network():
+    """Peak-targeted annealing shrinks both peak and op count on a
+    mid-size circuit (the r02 sweep: peak -65%, op -41% on this seed)."""
+    from tnc_amd.treesa import reduce_peak
+
+    tn = random_circuit(20, 12, 0.5, 0.7, 3, ConnectivityLayout.EAGLE)
+    g = Greedy().find_path(tn)
+    leaves = [LeafTensor(t.legs, t.bond_dims) for t in tn.tensors]
+    top, op, peak = reduce_peak(leaves, list(g.replace_path().toplevel), 0,
+                                moves=50_000, seed=1,
+                                initial_temperature=0.05,
+                                final_temperature=0.005)
+    assert peak < 0.5 * g.size
+    assert op < g.flops
+    validate_path(ContractionPath.simple(top))
+
+
+def test_cap_rescue_feasible_partition_paths():
+    """make_plan's local paths stay device-feasible at small trial counts
+    (the cap-rescue sweep in RandomGreedy): regression test for the r02
+    OOM where a 4-trial path peaked at 1.8e10 elements (293 GB)."""
+    from tnc_amd.contraction_path import flatten_network
+    from tnc_amd.dist import make_plan
+    from tnc_amd.executor import arena_bytes, plan_steps
+
+    tn, rp, meta = load_fixture("rqc36")
+    plan = make_plan(tn, 2, trials=4, size_cap=2.0e9)
+    for part in range(plan.nparts):
+        sub = plan.partitioned.tensors[part]
+        inner = plan.path.nested.get(part)
+        leaves, steps, _ = flatten_network(sub, inner)
+        ab = arena_bytes(leaves, steps, plan_steps(leaves, steps))
+        assert ab < 200e9, f"partition {part} arena {ab/1e9:.0f} GB"

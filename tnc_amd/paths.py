@@ -201,13 +201,9 @@ class RandomGreedy(_CotengrustLike):
         rng = np.random.Generator(np.random.PCG64(self.seed))
         best_path = None
         best_key = None
-        # trial 0 is the deterministic greedy; keeps RandomGreedy >= Greedy
-        for trial in range(max(1, self.ntrials)):
-            if trial == 0:
-                costmod, temperature = 1.0, 0.0
-            else:
-                costmod = math.exp(rng.uniform(math.log(0.1), math.log(4.0)))
-                temperature = math.exp(rng.uniform(math.log(0.001), math.log(1.0)))
+
+        def consider(costmod, temperature):
+            nonlocal best_path, best_key
             ssa = _greedy_ssa(leaves, costmod, temperature, rng)
             cost, peak = _ssa_op_cost(leaves, ssa)
             if self.size_cap is not None:
@@ -217,6 +213,31 @@ class RandomGreedy(_CotengrustLike):
             if best_key is None or key < best_key:
                 best_key = key
                 best_path = ssa
+
+        # trial 0 is the deterministic greedy; keeps RandomGreedy >= Greedy
+        for trial in range(max(1, self.ntrials)):
+            if trial == 0:
+                costmod, temperature = 1.0, 0.0
+            else:
+                costmod = math.exp(rng.uniform(math.log(0.1), math.log(4.0)))
+                temperature = math.exp(rng.uniform(math.log(0.001), math.log(1.0)))
+            consider(costmod, temperature)
+        # cap rescue: when no trial meets the size cap, keep sampling
+        # (alternating the low-costmod range, whose score minimizes
+        # size(ab) directly, with the standard range) until one fits or
+        # the extra budget runs out — without this, the "smallest peak
+        # wins" fallback can hand back paths whose peak exceeds device
+        # memory on partitioned networks
+        if self.size_cap is not None and best_key is not None and best_key[0]:
+            ranges = ((0.01, 0.5), (0.1, 4.0))
+            for extra in range(max(12, self.ntrials)):
+                lo, hi = ranges[extra % 2]
+                costmod = math.exp(rng.uniform(math.log(lo), math.log(hi)))
+                temperature = math.exp(
+                    rng.uniform(math.log(0.001), math.log(1.0)))
+                consider(costmod, temperature)
+                if not best_key[0]:
+                    break
         return best_path or []
 
 

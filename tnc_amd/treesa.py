@@ -220,6 +220,64 @@ def refine_replace_path(leaves: List[LeafTensor], toplevel, moves=100_000,
     return best[2], best[0], best[1]
 
 
+def reduce_peak(leaves: List[LeafTensor], toplevel, target_peak,
+                moves=100_000, seed=0, initial_temperature=1.0,
+                final_temperature=0.02):
+    """NNI anneal whose PRIMARY objective is the walk's peak step size
+    (out+a+b elements): memory-feasibility rescue for paths whose peak
+    exceeds device (or cap) limits. Secondary objective: op count.
+
+    Returns (toplevel, op, peak) of the best (peak, op)-lexicographic tree
+    seen. Stops early once peak <= target_peak and a further 25% of the
+    move budget brings no improvement."""
+    tree = _Tree(leaves, toplevel)
+    rng = np.random.default_rng(seed)
+    n = tree.n
+    if n < 3:
+        return list(toplevel), tree.total_op(), tree.peak()
+    internal = list(range(n, 2 * n - 1))
+    cur_op = tree.total_op()
+    cur_peak = tree.peak()
+    best = (cur_peak, cur_op, list(toplevel))
+    log_t0, log_t1 = math.log(initial_temperature), math.log(final_temperature)
+    stale = 0
+    for m in range(moves):
+        temperature = math.exp(log_t0 + (log_t1 - log_t0) * (m / moves))
+        x = internal[int(rng.integers(0, len(internal)))]
+        cs = int(rng.integers(0, 2))
+        gs = int(rng.integers(0, 2))
+        res = tree.nni(x, cs, gs, apply=False)
+        if res is None:
+            continue
+        delta, new_local_peak = res
+        if new_local_peak > cur_peak:
+            # peak-raising move: SA on the peak ratio
+            diff = math.log2(new_local_peak / cur_peak)
+            accept = math.exp(min(50.0, -diff / temperature)) >= rng.random()
+        elif delta <= 0:
+            accept = True
+        else:
+            diff = math.log2((cur_op + delta) / cur_op)
+            accept = math.exp(min(50.0, -diff / temperature)) >= rng.random()
+        if not accept:
+            continue
+        p = tree.left[x] if cs == 0 else tree.right[x]
+        old_p_peak, old_x_peak = tree.step_peak[p], tree.step_peak[x]
+        tree.nni(x, cs, gs, apply=True)
+        cur_op += delta
+        if new_local_peak >= cur_peak:
+            cur_peak = new_local_peak
+        elif max(old_p_peak, old_x_peak) >= cur_peak:
+            cur_peak = tree.peak()  # the argmax may have moved: recompute
+        stale += 1
+        if (cur_peak, cur_op) < (best[0], best[1]):
+            best = (cur_peak, cur_op, tree.to_replace_toplevel())
+            stale = 0
+        if best[0] <= target_peak and stale > moves // 4:
+            break
+    return best[2], best[1], best[0]
+
+
 class TreeSA:
     """Pathfinder: PartitionSearch (or a given base finder) followed by NNI
     tree annealing. find_path mirrors the Pathfinder trait (paths.rs:21-28);
