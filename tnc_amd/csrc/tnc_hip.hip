@@ -1181,6 +1181,25 @@ static bool permute_tiled(const CT* src, CT* dst, u64 elems,
   return true;
 }
 
+// Launch the best permute kernel for a gather described by `ax` (tiled
+// bit-permutation when applicable, else the index-gather permute).
+template <typename CT>
+static int launch_permute(const CT* src, CT* dst, u64 elems,
+                          const std::vector<AxisInfo>& ax,
+                          hipStream_t stream) {
+  if (!permute_tiled(src, dst, elems, ax, stream)) {
+    GatherMap map;
+    if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
+    int blocks = grid_for(elems);
+    if (map.pow2)
+      k_permute_ct<true><<<blocks, 256, 0, stream>>>(src, dst, elems, map);
+    else
+      k_permute_ct<false><<<blocks, 256, 0, stream>>>(src, dst, elems, map);
+  }
+  HIP_CHECK(hipGetLastError());
+  return TN_OK;
+}
+
 // core einsum over device buffers; out is contiguous row-major in out order.
 template <typename CT>
 static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
@@ -1454,8 +1473,6 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
   if (!is_ready(A, a_axes)) {
     std::vector<AxisInfo> ax;
     for (int axis : a_axes) ax.push_back({A.dims[axis], A.strides[axis], 0});
-    GatherMap map;
-    if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = M * K;
     {
       int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(CT));
@@ -1464,22 +1481,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       if (rc_ == TN_ERR_OOM && gather_ok) return run_gather();
       if (rc_) return rc_;
     }
-    if (!permute_tiled(Adata, packA, elems, ax, stream)) {
-      int blocks = grid_for(elems);
-      if (map.pow2)
-        k_permute_ct<true><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
-                                                       map);
-      else
-        k_permute_ct<false><<<blocks, 256, 0, stream>>>(Adata, packA, elems,
-                                                        map);
+    {
+      int rc_ = launch_permute(Adata, packA, elems, ax, stream);
+      if (rc_) return rc_;
     }
     Ag = packA;
   }
   if (!is_ready(B, b_axes)) {
     std::vector<AxisInfo> ax;
     for (int axis : b_axes) ax.push_back({B.dims[axis], B.strides[axis], 0});
-    GatherMap map;
-    if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
     u64 elems = K * N;
     {
       int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(CT));
@@ -1489,14 +1499,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       }
       if (rc_) return rc_;
     }
-    if (!permute_tiled(Bdata, packB, elems, ax, stream)) {
-      int blocks = grid_for(elems);
-      if (map.pow2)
-        k_permute_ct<true><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
-                                                       map);
-      else
-        k_permute_ct<false><<<blocks, 256, 0, stream>>>(Bdata, packB, elems,
-                                                        map);
+    {
+      int rc_ = launch_permute(Bdata, packB, elems, ax, stream);
+      if (rc_) return rc_;
     }
     Bg = packB;
   }
@@ -1606,18 +1611,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     std::vector<AxisInfo> ax;
     for (int i = 0; i < out_nd; ++i)
       ax.push_back({out_shape[i], tmp_stride[i], 0});
-    if (!permute_tiled((const CT*)tmpC, out, nout, ax, stream)) {
-      GatherMap map;
-      if (build_map(ax, &map)) FAILV(TN_ERR_INVALID, "rank too large");
-      int blocks = grid_for(nout);
-      if (map.pow2)
-        k_permute_ct<true><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
-                                                       nout, map);
-      else
-        k_permute_ct<false><<<blocks, 256, 0, stream>>>((const CT*)tmpC, out,
-                                                        nout, map);
+    {
+      int rc_ = launch_permute((const CT*)tmpC, out, nout, ax, stream);
+      if (rc_) return rc_;
     }
-    HIP_CHECK(hipGetLastError());
   }
   ws_free(ws, packA);
   ws_free(ws, packB);
@@ -1810,6 +1807,7 @@ struct tn_net {
   int dtype = 0;   // 0 = c128, 1 = c64
   size_t esize = 16;
   hipStream_t stream = nullptr;
+  hipStream_t stream2 = nullptr;  // pack-overlap stream (prepack permutes)
   std::vector<DevTensor> leaves;
   DevTensor final_t;       // final tensor of the last contract (owned)
   bool final_in_arena = false;
@@ -1853,6 +1851,12 @@ extern "C" tn_net* tn_net_create2(int device, int dtype) {
   net->dtype = dtype;
   net->esize = dtype == 0 ? 16 : 8;
   if (hipStreamCreate(&net->stream) != hipSuccess) {
+    delete net;
+    g_last_error = "hipStreamCreate failed";
+    return nullptr;
+  }
+  if (hipStreamCreate(&net->stream2) != hipSuccess) {
+    (void)hipStreamDestroy(net->stream);
     delete net;
     g_last_error = "hipStreamCreate failed";
     return nullptr;
@@ -1947,6 +1951,107 @@ static void symdiff(const DevTensor& a, const DevTensor& b,
   }
 }
 
+// ---- pack-overlap planning ------------------------------------------------
+// Mirror of einsum_dev_impl's dispatch, host-side and launch-free: would the
+// step take the TTGT route, and which operands would it pack? Used by the
+// walk to pre-permute the NEXT step's GEMM operands on a second stream while
+// the current step's GEMM runs (the GEMMs are MFMA-bound and use <10% of the
+// HBM bandwidth the permutes need, so the packs ride along ~free). Any
+// divergence from the real dispatch is performance-only: a prepacked operand
+// is a valid contiguous tensor in pack order, every kernel class consumes it
+// correctly, and an operand NOT prepacked is packed inline as before.
+struct PackPlan {
+  bool packA = false, packB = false;
+  std::vector<int> a_axes, b_axes;  // axis orders of the packed layouts
+  u64 MK = 0, KN = 0;               // packed element counts
+};
+
+static bool plan_prepack(const Meta& A, const Meta& B, const u64* out_labels,
+                         const u64* out_shape, int out_nd, PackPlan* plan) {
+  if (A.nd > TN_MAXR || B.nd > TN_MAXR || out_nd > TN_MAXR) return false;
+  auto find = [](const Meta& t, u64 lab) {
+    for (int i = 0; i < t.nd; ++i)
+      if (t.labels[i] == lab) return i;
+    return -1;
+  };
+  int apos[TN_MAXR], bpos[TN_MAXR];
+  u64 M = 1, N = 1, K = 1;
+  std::vector<int> m_out, n_out;
+  for (int i = 0; i < out_nd; ++i) {
+    apos[i] = find(A, out_labels[i]);
+    bpos[i] = find(B, out_labels[i]);
+    if (apos[i] >= 0 && bpos[i] >= 0) return false;
+    if (apos[i] >= 0) {
+      M *= out_shape[i];
+      m_out.push_back(i);
+    } else if (bpos[i] >= 0) {
+      N *= out_shape[i];
+      n_out.push_back(i);
+    } else {
+      return false;
+    }
+  }
+  std::vector<int> k_a, k_b;
+  for (int i = 0; i < A.nd; ++i) {
+    int j = find(B, A.labels[i]);
+    if (j >= 0) {
+      K *= A.dims[i];
+      k_a.push_back(i);
+      k_b.push_back(j);
+    }
+  }
+  if (M == 1 && N == 1 && K > TN_SMALLK) return false;  // dot route
+  bool skinny = (M < 16 || N < 16);
+  bool gemm_worthy = (K >= 16 && M >= MF_T && N >= MF_TN);
+  bool gather_ok = (K <= TN_SMALLK || skinny);
+  if (gather_ok && !gemm_worthy) return false;  // gather route
+  std::vector<int> a_axes, b_axes;
+  for (int p : m_out) a_axes.push_back(apos[p]);
+  for (int i : k_a) a_axes.push_back(i);
+  for (int j : k_b) b_axes.push_back(j);
+  for (int p : n_out) b_axes.push_back(bpos[p]);
+  auto is_ready = [](const Meta& t, const std::vector<int>& axes) {
+    if ((int)axes.size() != t.nd) return false;
+    i64 stride = 1;
+    for (int i = (int)axes.size() - 1; i >= 0; --i) {
+      if (t.strides[axes[i]] != stride) return false;
+      stride *= (i64)t.dims[axes[i]];
+    }
+    return true;
+  };
+  plan->packA = !is_ready(A, a_axes);
+  plan->packB = !is_ready(B, b_axes);
+  plan->a_axes = std::move(a_axes);
+  plan->b_axes = std::move(b_axes);
+  plan->MK = M * K;
+  plan->KN = K * N;
+  return plan->packA || plan->packB;
+}
+
+// fill a Meta from a contiguous row-major DevTensor
+static void meta_of(const DevTensor& t, Meta* m) {
+  m->nd = (int)t.labels.size();
+  for (int x = 0; x < m->nd; ++x) {
+    m->labels[x] = t.labels[x];
+    m->dims[x] = t.dims[x];
+  }
+  i64 stride = 1;
+  for (int x = m->nd - 1; x >= 0; --x) {
+    m->strides[x] = stride;
+    stride *= (i64)m->dims[x];
+  }
+  m->data = t.data;
+}
+
+static bool prepack_disabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TN_NO_PREPACK");
+    v = (e && e[0] && e[0] != '0') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
                          double* step_ms, double* gemm_ms, int32_t* kind,
                          double* elapsed_ms, bool capture = false,
@@ -1989,12 +2094,43 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
   }
 
   int rc = TN_OK;
+  // pack-overlap state: while step s runs on net->stream, the NEXT step's
+  // GEMM pack permutes run on net->stream2 (after an event marking steps
+  // < s complete). The replaced originals are freed at the top of step s+1,
+  // after the main stream waits on the packs — so every arena block's
+  // next user is ordered behind its last reader, preserving the arena's
+  // stream-ordered reuse contract across both streams.
+  struct Pending {
+    void* old_data;
+    bool old_owned;
+  };
+  std::vector<Pending> pending;
+  hipEvent_t pending_done = nullptr;
+  std::vector<hipEvent_t> prep_events;
+  const bool overlap =
+      !prepack_disabled() && net->stream2 && net->arena.base != nullptr;
+
   for (size_t s = 0; s < nsteps; ++s) {
     u64 i = pairs[2 * s], j = pairs[2 * s + 1];
     if (i >= n || j >= n || !alive[i] || !alive[j] || i == j) {
       rc = TN_ERR_INVALID;
       g_last_error = "invalid contraction path step";
       break;
+    }
+    // join this step's prepack (issued during step s-1): the packed
+    // operands replace slots[i]/slots[j]; free the originals only now,
+    // behind the wait, so their blocks cannot be re-handed while the pack
+    // kernels still read them
+    if (pending_done) {
+      if (hipStreamWaitEvent(net->stream, pending_done, 0) != hipSuccess) {
+        rc = TN_ERR_HIP;
+        g_last_error = "hipStreamWaitEvent failed (prepack join)";
+        break;
+      }
+      for (auto& p : pending)
+        if (p.old_owned && p.old_data) ws_free(ws, p.old_data);
+      pending.clear();
+      pending_done = nullptr;
     }
     DevTensor& A = slots[i];
     DevTensor& B = slots[j];
@@ -2008,29 +2144,112 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
       break;
     }
     out.owned = true;
+
+    // prepack the next step's GEMM operands on stream2 (overlaps this
+    // step's kernels). A prepacked slot becomes a contiguous tensor in
+    // pack order — einsum_dev_impl then sees is_ready and skips its
+    // inline pack. Arena-only; on any shortage the step simply packs
+    // inline as before.
+    if (overlap && s + 1 < nsteps) {
+      u64 ni = pairs[2 * (s + 1)], nj = pairs[2 * (s + 1) + 1];
+      bool valid = ni < n && nj < n && ni != nj && ni != j && nj != j &&
+                   (ni == i || alive[ni]) && (nj == i || alive[nj]);
+      if (valid) {
+        const DevTensor& TA = (ni == i) ? out : slots[ni];
+        const DevTensor& TB = (nj == i) ? out : slots[nj];
+        std::vector<u64> nlabels, ndims;
+        symdiff(TA, TB, &nlabels, &ndims);
+        Meta mna, mnb;
+        PackPlan plan;
+        if ((int)TA.labels.size() <= TN_MAXR &&
+            (int)TB.labels.size() <= TN_MAXR &&
+            (int)nlabels.size() <= TN_MAXR) {
+          meta_of(TA, &mna);
+          meta_of(TB, &mnb);
+          if (plan_prepack(mna, mnb, nlabels.data(), ndims.data(),
+                           (int)nlabels.size(), &plan)) {
+            void* dstA = nullptr;
+            void* dstB = nullptr;
+            bool wantA = plan.packA && ni != i;
+            bool wantB = plan.packB && nj != i;
+            if (wantA) dstA = net->arena.alloc(plan.MK * net->esize);
+            if (wantB) dstB = net->arena.alloc(plan.KN * net->esize);
+            if ((wantA && !dstA) || (wantB && !dstB)) {
+              // partial shortage: keep it simple, pack inline at s+1
+              if (dstA) net->arena.release(dstA);
+              if (dstB) net->arena.release(dstB);
+              dstA = dstB = nullptr;
+            }
+            if (dstA || dstB) {
+              hipEvent_t e_ready = nullptr, e_done = nullptr;
+              if (hipEventCreate(&e_ready) != hipSuccess) e_ready = nullptr;
+              if (e_ready && hipEventCreate(&e_done) != hipSuccess) {
+                (void)hipEventDestroy(e_ready);
+                e_ready = nullptr;
+              }
+              if (!e_ready) {
+                if (dstA) net->arena.release(dstA);
+                if (dstB) net->arena.release(dstB);
+              } else {
+                prep_events.push_back(e_ready);
+                prep_events.push_back(e_done);
+                if (hipEventRecord(e_ready, net->stream) != hipSuccess ||
+                    hipStreamWaitEvent(net->stream2, e_ready, 0) !=
+                        hipSuccess) {
+                  rc = TN_ERR_HIP;
+                  g_last_error = "prepack event sync failed";
+                  break;
+                }
+                auto do_pack = [&](DevTensor& S, void* dst,
+                                   const std::vector<int>& axes,
+                                   u64 elems) -> int {
+                  Meta ms;
+                  meta_of(S, &ms);
+                  std::vector<AxisInfo> ax;
+                  for (int axis : axes)
+                    ax.push_back({ms.dims[axis], ms.strides[axis], 0});
+                  int prc =
+                      net->dtype == 0
+                          ? launch_permute((const double2*)S.data,
+                                           (double2*)dst, elems, ax,
+                                           net->stream2)
+                          : launch_permute((const float2*)S.data,
+                                           (float2*)dst, elems, ax,
+                                           net->stream2);
+                  if (prc != TN_OK) return prc;
+                  pending.push_back({S.owned ? S.data : nullptr, S.owned});
+                  std::vector<u64> nl, nd;
+                  for (int axis : axes) {
+                    nl.push_back(S.labels[axis]);
+                    nd.push_back(S.dims[axis]);
+                  }
+                  S.labels = std::move(nl);
+                  S.dims = std::move(nd);
+                  S.data = dst;
+                  S.owned = true;
+                  S.external = false;
+                  return TN_OK;
+                };
+                if (dstA) rc = do_pack(slots[ni], dstA, plan.a_axes, plan.MK);
+                if (rc == TN_OK && dstB)
+                  rc = do_pack(slots[nj], dstB, plan.b_axes, plan.KN);
+                if (rc != TN_OK) break;
+                if (hipEventRecord(e_done, net->stream2) != hipSuccess) {
+                  rc = TN_ERR_HIP;
+                  g_last_error = "prepack event record failed";
+                  break;
+                }
+                pending_done = e_done;
+              }
+            }
+          }
+        }
+      }
+    }
+
     Meta ma, mb;
-    ma.nd = (int)A.labels.size();
-    for (int x = 0; x < ma.nd; ++x) {
-      ma.labels[x] = A.labels[x];
-      ma.dims[x] = A.dims[x];
-    }
-    i64 stride = 1;
-    for (int x = ma.nd - 1; x >= 0; --x) {
-      ma.strides[x] = stride;
-      stride *= (i64)ma.dims[x];
-    }
-    ma.data = A.data;
-    mb.nd = (int)B.labels.size();
-    for (int x = 0; x < mb.nd; ++x) {
-      mb.labels[x] = B.labels[x];
-      mb.dims[x] = B.dims[x];
-    }
-    stride = 1;
-    for (int x = mb.nd - 1; x >= 0; --x) {
-      mb.strides[x] = stride;
-      stride *= (i64)mb.dims[x];
-    }
-    mb.data = B.data;
+    meta_of(A, &ma);
+    meta_of(B, &mb);
 
     if (profiled) HIP_CHECK(hipEventRecord(ev[4 * s], net->stream));
     StepStats st;
@@ -2127,16 +2346,23 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
     }
   }
   if (rc != TN_OK) {
-    // free any owned intermediates left over
+    // free any owned intermediates left over, plus originals whose
+    // prepack-deferred frees never ran
+    if (!capture) {
+      (void)hipStreamSynchronize(net->stream);
+      (void)hipStreamSynchronize(net->stream2);
+    }
+    for (auto& p : pending)
+      if (p.old_owned && p.old_data) ws_free(ws, p.old_data);
     for (size_t x = 0; x < n; ++x)
       if (alive[x] && slots[x].owned && slots[x].data)
         ws_free(ws, slots[x].data);
-    if (!capture) (void)hipStreamSynchronize(net->stream);
   }
   if (spilled) *spilled = ws.spilled;
   if (walk_start) (void)hipEventDestroy(walk_start);
   if (walk_end) (void)hipEventDestroy(walk_end);
   for (auto& e : ev) (void)hipEventDestroy(e);
+  for (auto& e : prep_events) (void)hipEventDestroy(e);
   return rc;
 }
 
@@ -2314,5 +2540,6 @@ extern "C" void tn_net_destroy(tn_net* net) {
     (void)hipFree(net->final_t.data);
   net->arena.destroy();
   (void)hipStreamDestroy(net->stream);
+  if (net->stream2) (void)hipStreamDestroy(net->stream2);
   delete net;
 }

@@ -63,16 +63,11 @@ def plan_steps(leaves, steps):
     return infos
 
 
-def _step_ws_bytes(info, esize):
-    """Workspace the C dispatch will allocate for one step (mirrors
-    einsum_dev_impl: dot partials, nothing for gather kernels, pack
-    buffers for TTGT). Keep in sync with TN_SMALLK=64, MF_T=128,
-    MF_TN=64 in tnc_amd/csrc/tnc_hip.hip."""
-    if info.m == 1 and info.n == 1 and info.k > 64:
-        return 1 << 25  # dot partial buffer (<= 2^21 blocks * 16 B)
+def _pack_ws_bytes(info, esize):
+    """Pack-permute workspace of a TTGT step (0 for dot/gather routes)."""
     gemm_worthy = info.k >= 16 and info.m >= 128 and info.n >= 64
     if (info.k <= 64 or info.m < 16 or info.n < 16) and not gemm_worthy:
-        return 0  # smallk/anyk gather: no workspace
+        return 0
     ws = 0
     if info.packa:
         ws += info.m * info.k * esize
@@ -81,15 +76,29 @@ def _step_ws_bytes(info, esize):
     return ws
 
 
+def _step_ws_bytes(info, esize):
+    """Workspace the C dispatch will allocate for one step (mirrors
+    einsum_dev_impl: dot partials, nothing for gather kernels, pack
+    buffers for TTGT). Keep in sync with TN_SMALLK=64, MF_T=128,
+    MF_TN=64 in tnc_amd/csrc/tnc_hip.hip."""
+    if info.m == 1 and info.n == 1 and info.k > 64:
+        return 1 << 25  # dot partial buffer (<= 2^21 blocks * 16 B)
+    return _pack_ws_bytes(info, esize)
+
+
 def arena_bytes(leaves, steps, infos, esize=16):
     """Peak device-arena demand: live intermediates + this step's output and
-    the workspaces its dispatch actually uses, walked over the plan.
+    the workspaces its dispatch actually uses, walked over the plan. The
+    pack-overlap executor allocates the NEXT step's pack buffers one step
+    early (prepack on stream2), so each step's demand includes them.
     Padded 15% for fragmentation + split-K slack."""
     live = {}  # slot -> bytes (intermediates only; leaves live outside)
     peak = 0
-    for info in infos:
+    for s, info in enumerate(infos):
         out_b = info.m * info.n * esize
-        demand = sum(live.values()) + out_b + _step_ws_bytes(info, esize)
+        nxt = _pack_ws_bytes(infos[s + 1], esize) if s + 1 < len(infos) else 0
+        demand = (sum(live.values()) + out_b + _step_ws_bytes(info, esize)
+                  + nxt)
         peak = max(peak, demand)
         live.pop(info.j, None)
         live[info.i] = out_b
