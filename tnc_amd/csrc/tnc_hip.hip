@@ -820,6 +820,131 @@ __global__ __launch_bounds__(MF_THREADS) void k_zgemm_c128_glds_pure(
 }
 
 
+// Gather-staged c128 glds GEMM: reads A and B straight through the TTGT
+// pack PERMUTATION instead of from pre-packed copies — the pack kernels
+// (and their read+write HBM traffic, ~54 GB per rqc36 contraction)
+// disappear. Works because with pow2 dims the packed offset is SEPARABLE:
+// src = rowOff(m) + khi(k0) + klo(c), where rowOff/klo are tiny per-block
+// LDS tables built in the prologue and khi is a wave-uniform (SALU) sum
+// per k-iteration; the per-lane staging address is then two adds, no
+// heavier than the multiply it replaces. Everything else (XOR-swizzled
+// interleaved LDS image, MFMA body, epilogue) is identical to
+// k_zgemm_c128_glds_pure. Launched only for pure shapes (M%128 == 0,
+// N%64 == 0, K%16 == 0) with all-pow2 dims.
+struct GatherGemmMap {
+  int rbits, kbits;   // bits of the row index (M for A / N for B) and of k
+  u64 rstride[34];    // source stride (elements) contributed by row bit b
+  u64 kstride[34];    // source stride contributed by k bit b
+};
+
+__global__ __launch_bounds__(MF_THREADS) void k_zgemm_c128_glds_gather(
+    const double2* __restrict__ A, const double2* __restrict__ B,
+    double2* __restrict__ C, u64 M, u64 N, u64 K, unsigned col_tiles,
+    unsigned tiles, u64 kchunk, GatherGemmMap mA, GatherGemmMap mB) {
+  constexpr int TM = MF_T, TN = MF_TN, KT = MF_K;
+  __shared__ double2 As[TM * KT];
+  __shared__ double2 Bs[KT * TN];
+  __shared__ u64 rowOffA[TM];
+  __shared__ u64 colOffB[TN];
+  __shared__ u64 kloA[KT];
+  __shared__ u64 kloB[KT];
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const unsigned tile = (unsigned)blockIdx.x % tiles;
+  const unsigned slice = (unsigned)blockIdx.x / tiles;
+  const u64 brow = (u64)(tile / col_tiles) * TM;
+  const u64 bcol = (u64)(tile % col_tiles) * TN;
+  const u64 kbeg = (u64)slice * kchunk;
+  const u64 kend = (kbeg + kchunk < K) ? kbeg + kchunk : K;
+  C += (u64)slice * M * N;
+  // offset tables (once per block)
+  for (int t = threadIdx.x; t < TM + TN + 2 * KT; t += MF_THREADS) {
+    if (t < TM) {
+      u64 idx = brow + t, off = 0;
+      for (int b = 0; b < mA.rbits; ++b)
+        if ((idx >> b) & 1) off += mA.rstride[b];
+      rowOffA[t] = off;
+    } else if (t < TM + TN) {
+      u64 idx = bcol + (t - TM), off = 0;
+      for (int b = 0; b < mB.rbits; ++b)
+        if ((idx >> b) & 1) off += mB.rstride[b];
+      colOffB[t - TM] = off;
+    } else if (t < TM + TN + KT) {
+      int c = t - TM - TN;
+      u64 off = 0;
+      for (int b = 0; b < 4; ++b)
+        if ((c >> b) & 1) off += mA.kstride[b];
+      kloA[c] = off;
+    } else {
+      int c = t - TM - TN - KT;
+      u64 off = 0;
+      for (int b = 0; b < 4; ++b)
+        if ((c >> b) & 1) off += mB.kstride[b];
+      kloB[c] = off;
+    }
+  }
+  __syncthreads();
+  v4d cr[4], ci[4];
+  for (int f = 0; f < 4; ++f) {
+    cr[f] = v4d{0, 0, 0, 0};
+    ci[f] = v4d{0, 0, 0, 0};
+  }
+  const int fi = lane % 16;
+  for (u64 k0 = kbeg; k0 < kend; k0 += KT) {
+    u64 khiA = 0, khiB = 0;
+    for (int b = 4; b < mA.kbits; ++b)
+      if ((k0 >> b) & 1) khiA += mA.kstride[b];
+    for (int b = 4; b < mB.kbits; ++b)
+      if ((k0 >> b) & 1) khiB += mB.kstride[b];
+    const double2* Ak = A + khiA;
+    const double2* Bk = B + khiB;
+    for (int piece = 0; piece < 4; ++piece) {
+      int base = (wave * 4 + piece) * 64;
+      int i = base + lane;
+      int r = i / KT, c_sw = i % KT;
+      int c = c_sw ^ (r & 15);
+      const double2* src = &Ak[rowOffA[r] + kloA[c]];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&As[base], 16, 0, 0);
+    }
+    for (int piece = 0; piece < 2; ++piece) {
+      int base = piece * 512 + wave * 64;
+      int j = base + lane;
+      int k = j / TN, col_sw = j % TN;
+      int col = col_sw ^ ((k & 3) << 4);
+      const double2* src = &Bk[kloB[k] + colOffB[col]];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&Bs[base], 16, 0, 0);
+    }
+    __syncthreads();
+    for (int kq = 0; kq < KT / 4; ++kq) {
+      const int arow = wave * 16 + fi;
+      const int ak = kq * 4 + (lane / 16);
+      double2 a = As[arow * KT + (ak ^ (arow & 15))];
+      for (int f = 0; f < 4; ++f) {
+        const int bcolf = f * 16 + fi;
+        double2 b = Bs[ak * TN + (bcolf ^ ((ak & 3) << 4))];
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.x, cr[f], 0, 0, 0);
+        cr[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(-a.y, b.y, cr[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.x, b.y, ci[f], 0, 0, 0);
+        ci[f] = __builtin_amdgcn_mfma_f64_16x16x4f64(a.y, b.x, ci[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow0 = wave * 16 + (lane / 16);
+  const int ccol = lane % 16;
+  for (int f = 0; f < 4; ++f)
+    for (int r = 0; r < 4; ++r) {
+      u64 row = brow + crow0 + 4 * r;
+      u64 col = bcol + f * 16 + ccol;
+      C[row * N + col] = make_double2(cr[f][r], ci[f][r]);
+    }
+}
+
+
 // c64 pure-glds kernel: one 16-byte LDS-DMA moves TWO float2 elements, so
 // the swizzle works at pair granularity (8 pairs per 16-deep K row).
 __global__ __launch_bounds__(MF_THREADS) void k_zgemm_c64_glds_pure(
@@ -1181,6 +1306,39 @@ static bool permute_tiled(const CT* src, CT* dst, u64 elems,
   return true;
 }
 
+// Build the bit-scatter map of a packed [row][k] (or [k][col]) layout:
+// packed-index bit b -> source stride. Axes are outer..inner; the inner
+// axis supplies the low bits. Fails on non-pow2 dims or non-positive
+// strides (the gather GEMM then falls back to the pack path).
+static bool build_ggmap(const Meta& t, const std::vector<int>& axes_r,
+                        const std::vector<int>& axes_k, GatherGemmMap* m) {
+  auto fill = [&](const std::vector<int>& axes, u64* stride, int* nbits) {
+    int nb = 0;
+    for (int x = (int)axes.size() - 1; x >= 0; --x) {
+      u64 d = t.dims[axes[x]];
+      if (d & (d - 1)) return false;
+      if (d > 1 && t.strides[axes[x]] <= 0) return false;
+      for (u64 v = 1; v < d; v <<= 1) {
+        if (nb >= 34) return false;
+        stride[nb++] = (u64)t.strides[axes[x]] * v;
+      }
+    }
+    *nbits = nb;
+    return true;
+  };
+  return fill(axes_r, m->rstride, &m->rbits) &&
+         fill(axes_k, m->kstride, &m->kbits);
+}
+
+static bool gather_gemm_disabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TN_NO_GATHER_GEMM");
+    v = (e && e[0] && e[0] != '0') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 // Launch the best permute kernel for a gather described by `ax` (tiled
 // bit-permutation when applicable, else the index-gather permute).
 template <typename CT>
@@ -1466,11 +1624,30 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     return true;
   };
 
+  const bool needA = !is_ready(A, a_axes);
+  const bool needB = !is_ready(B, b_axes);
+  const bool mfma_shape = (M >= 32 && N >= 32);
+  // gather-staged GEMM (c128): fold the pack permutes into the GEMM's LDS
+  // staging when the shape is pure and every dim is pow2 — no pack
+  // kernels, no pack workspace, no pack HBM traffic
+  GatherGemmMap gmA{}, gmB{};
+  bool gather_gemm = false;
+  if constexpr (std::is_same_v<CT, double2>) {
+    if ((needA || needB) && mfma_shape && (M % MF_T == 0) &&
+        (N % MF_TN == 0) && (K % MF_K == 0) && !gather_gemm_disabled()) {
+      std::vector<int> a_m, b_n;
+      for (int p : m_out) a_m.push_back(apos[p]);
+      for (int p : n_out) b_n.push_back(bpos[p]);
+      gather_gemm = build_ggmap(A, a_m, k_a, &gmA) &&
+                    build_ggmap(B, b_n, k_b, &gmB);
+    }
+  }
+
   const CT* Ag = Adata;
   const CT* Bg = Bdata;
   CT* packA = nullptr;
   CT* packB = nullptr;
-  if (!is_ready(A, a_axes)) {
+  if (!gather_gemm && needA) {
     std::vector<AxisInfo> ax;
     for (int axis : a_axes) ax.push_back({A.dims[axis], A.strides[axis], 0});
     u64 elems = M * K;
@@ -1487,7 +1664,7 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     }
     Ag = packA;
   }
-  if (!is_ready(B, b_axes)) {
+  if (!gather_gemm && needB) {
     std::vector<AxisInfo> ax;
     for (int axis : b_axes) ax.push_back({B.dims[axis], B.strides[axis], 0});
     u64 elems = K * N;
@@ -1526,7 +1703,7 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     Cg = tmpC;
   }
 
-  bool mfma = (M >= 32 && N >= 32);
+  bool mfma = mfma_shape;
   u64 row_tile_h = mfma ? MF_T : GT;
   u64 row_tiles = (M + row_tile_h - 1) / row_tile_h;
   u64 col_tiles = (N + GT - 1) / GT;
@@ -1562,7 +1739,13 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     if constexpr (std::is_same_v<CT, double2>) {
       bool pure = (M % MF_T == 0) && (N % MF_TN == 0) && (K % MF_K == 0) &&
                   (kchunk % MF_K == 0);
-      if (pure)
+      // gather_gemm implies pure: its precondition covers M/N/K tiling and
+      // kchunk is MF_K-rounded above
+      if (gather_gemm)
+        k_zgemm_c128_glds_gather<<<grid, MF_THREADS, 0, stream>>>(
+            Adata, Bdata, gemm_out, M, N, K, (unsigned)col_tiles,
+            (unsigned)tiles, kchunk, gmA, gmB);
+      else if (pure)
         k_zgemm_c128_glds_pure<<<grid, MF_THREADS, 0, stream>>>(
             Ag, Bg, gemm_out, M, N, K, (unsigned)col_tiles, (unsigned)tiles,
             kchunk);
@@ -1967,7 +2150,8 @@ struct PackPlan {
 };
 
 static bool plan_prepack(const Meta& A, const Meta& B, const u64* out_labels,
-                         const u64* out_shape, int out_nd, PackPlan* plan) {
+                         const u64* out_shape, int out_nd, PackPlan* plan,
+                         bool c128 = false) {
   if (A.nd > TN_MAXR || B.nd > TN_MAXR || out_nd > TN_MAXR) return false;
   auto find = [](const Meta& t, u64 lab) {
     for (int i = 0; i < t.nd; ++i)
@@ -2005,6 +2189,20 @@ static bool plan_prepack(const Meta& A, const Meta& B, const u64* out_labels,
   bool gemm_worthy = (K >= 16 && M >= MF_T && N >= MF_TN);
   bool gather_ok = (K <= TN_SMALLK || skinny);
   if (gather_ok && !gemm_worthy) return false;  // gather route
+  // steps the gather-staged GEMM will take need no pack at all
+  if (c128 && (M >= 32 && N >= 32) && (M % MF_T == 0) && (N % MF_TN == 0) &&
+      (K % MF_K == 0) && !gather_gemm_disabled()) {
+    bool pow2 = true;
+    for (int i = 0; i < A.nd; ++i)
+      if ((A.dims[i] & (A.dims[i] - 1)) ||
+          (A.dims[i] > 1 && A.strides[i] <= 0))
+        pow2 = false;
+    for (int i = 0; i < B.nd; ++i)
+      if ((B.dims[i] & (B.dims[i] - 1)) ||
+          (B.dims[i] > 1 && B.strides[i] <= 0))
+        pow2 = false;
+    if (pow2) return false;
+  }
   std::vector<int> a_axes, b_axes;
   for (int p : m_out) a_axes.push_back(apos[p]);
   for (int i : k_a) a_axes.push_back(i);
@@ -2167,7 +2365,8 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
           meta_of(TA, &mna);
           meta_of(TB, &mnb);
           if (plan_prepack(mna, mnb, nlabels.data(), ndims.data(),
-                           (int)nlabels.size(), &plan)) {
+                           (int)nlabels.size(), &plan,
+                           net->dtype == 0)) {
             void* dstA = nullptr;
             void* dstB = nullptr;
             bool wantA = plan.packA && ni != i;
