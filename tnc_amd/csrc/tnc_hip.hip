@@ -1199,6 +1199,12 @@ struct WsCtx {
   bool capturing = false;
   bool spilled = false;
   std::vector<void*>* deferred = nullptr;
+  // pack-pipeline stream (null = pipelining off): pack permutes of the
+  // TTGT route may run K-window-chunked on this stream, overlapped with
+  // the window GEMMs on `stream`
+  hipStream_t stream2 = nullptr;
+  // deferred event destruction (capture mode); null = destroy immediately
+  std::vector<hipEvent_t>* events = nullptr;
 };
 
 // Non-arena workspace uses plain hipMalloc, NOT hipMallocAsync: on this
@@ -1330,11 +1336,27 @@ static bool build_ggmap(const Meta& t, const std::vector<int>& axes_r,
          fill(axes_k, m->kstride, &m->kbits);
 }
 
+static bool pipeline_disabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TN_NO_PIPELINE");
+    v = (e && e[0] && e[0] != '0') ? 1 : 0;
+  }
+  return v == 1;
+}
+
+// The gather-staged GEMM is OPT-IN (TN_GATHER_GEMM=1): measured on the
+// rqc36 dominant shapes it is ~35% SLOWER than pack + pure GEMM (44.9 vs
+// 69.6 TF/s on M16384/N4096/K32768) because the GEMM re-reads its operand
+// strips ~12x across tiles — the pack is a bandwidth AMORTIZER (scattered
+// source read once, packed copy re-read coalesced), not removable
+// overhead. Kept for low-reuse shapes and as a recorded negative result
+// (DESIGN.md "Round-2 measured state").
 static bool gather_gemm_disabled() {
   static int v = -1;
   if (v < 0) {
-    const char* e = getenv("TN_NO_GATHER_GEMM");
-    v = (e && e[0] && e[0] != '0') ? 1 : 0;
+    const char* e = getenv("TN_GATHER_GEMM");
+    v = (e && e[0] && e[0] != '0') ? 0 : 1;
   }
   return v == 1;
 }
@@ -1643,6 +1665,198 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     }
   }
 
+  // does C == out directly? (decided before packing so the pack-pipeline
+  // below can target the right C buffer)
+  bool direct = true;
+  if (!m_out.empty() && !n_out.empty() && m_out.back() > n_out.front())
+    direct = false;
+  CT* Cg = out;
+  CT* tmpC = nullptr;
+  if (!direct) {
+    if (stats) stats->kind = 3;
+    {
+      int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(CT));
+      if (rc_ == TN_ERR_OOM && gather_ok) return run_gather();
+      if (rc_) return rc_;
+    }
+    Cg = tmpC;
+  }
+
+  // ---- pack-pipeline: chunk the pack over K windows and overlap each
+  // window's permute (stream2) with the previous window's GEMM (stream).
+  // The window GEMMs write split-K-style slices reduced at the end, so no
+  // kernel changes are needed; the pack permutes (HBM-bound, ~4 TB/s)
+  // ride in the MFMA-bound GEMM's spare bandwidth. Applied when the model
+  // predicts a net win over the serial pack (the slice buffers cost
+  // (2p)·M·N extra traffic). Requires a packed A (a ready A's K-windows
+  // are strided); B may be packed or ready (row windows are contiguous).
+  bool pipelined = false;
+  if (ws.stream2 && needA && mfma_shape && (M % MF_T == 0) &&
+      (N % MF_TN == 0) && (K % MF_K == 0) && !pipeline_disabled()) {
+    u64 packbytes =
+        ((needA ? M * K : 0) + (needB ? K * N : 0)) * sizeof(CT);
+    double best_save = 0.0;
+    int best_p = 0, best_npre = 0;
+    u64 best_kc = 0;
+    {
+      u64 p = 1;
+      u64 tiles_w = (M / MF_T) * ((N + MF_TN - 1) / MF_TN);
+      for (size_t x = 0; x < k_a.size() && p < 16; ++x) {
+        p *= A.dims[k_a[x]];
+        if (p < 2 || p > 16) continue;
+        u64 kc = K / p;
+        if (kc % MF_K || tiles_w < 256) continue;
+        double save = 2.0 * (double)packbytes / 4e12 * (1.0 - 1.0 / p) -
+                      2.0 * p * (double)(M * N * sizeof(CT)) / 6e12;
+        if (save > best_save) {
+          best_save = save;
+          best_p = (int)p;
+          best_kc = kc;
+          best_npre = (int)(x + 1);
+        }
+      }
+    }
+    if (best_p && best_save > 1e-4) {
+      const int P = best_p;
+      const u64 kc = best_kc;
+      CT* Awin = nullptr;
+      CT* Bwin = nullptr;
+      CT* slices = nullptr;
+      int rc_ = ws_alloc(ws, (void**)&Awin, M * K * sizeof(CT));
+      if (rc_ == TN_OK && needB)
+        rc_ = ws_alloc(ws, (void**)&Bwin, K * N * sizeof(CT));
+      if (rc_ == TN_OK)
+        rc_ = ws_alloc(ws, (void**)&slices, (u64)P * nout * sizeof(CT));
+      hipEvent_t e0 = nullptr, ep[16] = {};
+      bool ev_ok = (rc_ == TN_OK);
+      if (ev_ok && hipEventCreate(&e0) != hipSuccess) {
+        e0 = nullptr;
+        ev_ok = false;
+      }
+      for (int w = 0; ev_ok && w < P; ++w)
+        if (hipEventCreate(&ep[w]) != hipSuccess) {
+          ep[w] = nullptr;
+          ev_ok = false;
+        }
+      if (!ev_ok) {
+        // shortage: fall through to the serial pack path
+        ws_free(ws, Awin);
+        ws_free(ws, Bwin);
+        ws_free(ws, slices);
+        if (ws.events) {
+          if (e0) ws.events->push_back(e0);
+          for (int w = 0; w < P; ++w)
+            if (ep[w]) ws.events->push_back(ep[w]);
+        } else {
+          if (e0) (void)hipEventDestroy(e0);
+          for (int w = 0; w < P; ++w)
+            if (ep[w]) (void)hipEventDestroy(ep[w]);
+        }
+      } else {
+        // mixed-radix suffix products of the leading K legs (window digit
+        // x has radix A.dims[k_a[x]])
+        u64 sufA[16] = {}, sufB[16] = {};
+        {
+          u64 s = 1;
+          for (int x = best_npre - 1; x >= 0; --x) {
+            sufA[x] = s;
+            s *= A.dims[k_a[x]];
+          }
+        }
+        HIP_CHECK(hipEventRecord(e0, stream));
+        HIP_CHECK(hipStreamWaitEvent(ws.stream2, e0, 0));
+        for (int w = 0; w < P; ++w) {
+          // source offsets of window w
+          u64 offA = 0, offB = 0;
+          for (int x = 0; x < best_npre; ++x) {
+            u64 digit = ((u64)w / sufA[x]) % A.dims[k_a[x]];
+            offA += digit * (u64)A.strides[k_a[x]];
+            offB += digit * (u64)B.strides[k_b[x]];
+          }
+          {
+            std::vector<AxisInfo> ax;
+            for (int p2 : m_out)
+              ax.push_back({A.dims[apos[p2]], A.strides[apos[p2]], 0});
+            for (size_t x = best_npre; x < k_a.size(); ++x)
+              ax.push_back({A.dims[k_a[x]], A.strides[k_a[x]], 0});
+            int prc = launch_permute(Adata + offA, Awin + (u64)w * M * kc,
+                                     M * kc, ax, ws.stream2);
+            if (prc) return prc;
+          }
+          if (needB) {
+            std::vector<AxisInfo> ax;
+            for (size_t x = best_npre; x < k_b.size(); ++x)
+              ax.push_back({B.dims[k_b[x]], B.strides[k_b[x]], 0});
+            for (int p2 : n_out)
+              ax.push_back({B.dims[bpos[p2]], B.strides[bpos[p2]], 0});
+            int prc = launch_permute(Bdata + offB, Bwin + (u64)w * kc * N,
+                                     kc * N, ax, ws.stream2);
+            if (prc) return prc;
+          }
+          HIP_CHECK(hipEventRecord(ep[w], ws.stream2));
+        }
+        u64 tiles_w = (M / MF_T) * (N / MF_TN);
+        if (stats && stats->gemm_ev0)
+          HIP_CHECK(hipEventRecord(stats->gemm_ev0, stream));
+        for (int w = 0; w < P; ++w) {
+          HIP_CHECK(hipStreamWaitEvent(stream, ep[w], 0));
+          const CT* Aw = Awin + (u64)w * M * kc;
+          const CT* Bw = needB ? Bwin + (u64)w * kc * N
+                               : Bdata + (u64)w * kc * N;
+          CT* Cw = slices + (u64)w * nout;
+          if constexpr (std::is_same_v<CT, double2>) {
+            k_zgemm_c128_glds_pure<<<dim3((unsigned)tiles_w), MF_THREADS, 0,
+                                     stream>>>(
+                Aw, Bw, Cw, M, N, kc, (unsigned)(N / MF_TN),
+                (unsigned)tiles_w, kc);
+          } else {
+            k_zgemm_c64_glds_pure<<<dim3((unsigned)tiles_w), MF_THREADS, 0,
+                                    stream>>>(
+                (const float2*)Aw, (const float2*)Bw, (float2*)Cw, M, N, kc,
+                (unsigned)(N / MF_TN), (unsigned)tiles_w, kc);
+          }
+        }
+        k_splitk_reduce<<<grid_for(nout), 256, 0, stream>>>(slices, Cg,
+                                                            nout, P);
+        if (stats && stats->gemm_ev1)
+          HIP_CHECK(hipEventRecord(stats->gemm_ev1, stream));
+        HIP_CHECK(hipGetLastError());
+        ws_free(ws, Awin);
+        ws_free(ws, Bwin);
+        ws_free(ws, slices);
+        if (ws.events) {
+          ws.events->push_back(e0);
+          for (int w = 0; w < P; ++w) ws.events->push_back(ep[w]);
+        } else {
+          (void)hipEventDestroy(e0);
+          for (int w = 0; w < P; ++w) (void)hipEventDestroy(ep[w]);
+        }
+        pipelined = true;
+      }
+    }
+  }
+  if (pipelined) {
+    if (!direct) {
+      std::vector<i64> tmp_stride(out_nd, 0);
+      i64 stride = 1;
+      for (int t = (int)n_out.size() - 1; t >= 0; --t) {
+        tmp_stride[n_out[t]] = stride;
+        stride *= (i64)out_shape[n_out[t]];
+      }
+      for (int t = (int)m_out.size() - 1; t >= 0; --t) {
+        tmp_stride[m_out[t]] = stride;
+        stride *= (i64)out_shape[m_out[t]];
+      }
+      std::vector<AxisInfo> ax;
+      for (int i = 0; i < out_nd; ++i)
+        ax.push_back({out_shape[i], tmp_stride[i], 0});
+      int rc_ = launch_permute((const CT*)tmpC, out, nout, ax, stream);
+      if (rc_) return rc_;
+    }
+    ws_free(ws, tmpC);
+    return TN_OK;
+  }
+
   const CT* Ag = Adata;
   const CT* Bg = Bdata;
   CT* packA = nullptr;
@@ -1681,26 +1895,6 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       if (rc_) return rc_;
     }
     Bg = packB;
-  }
-
-  // does C == out directly?
-  bool direct = true;
-  if (!m_out.empty() && !n_out.empty() && m_out.back() > n_out.front())
-    direct = false;
-  CT* Cg = out;
-  CT* tmpC = nullptr;
-  if (!direct) {
-    if (stats) stats->kind = 3;
-    {
-      int rc_ = ws_alloc(ws, (void**)&tmpC, nout * sizeof(CT));
-      if (rc_ == TN_ERR_OOM && gather_ok) {
-        ws_free(ws, packA);
-        ws_free(ws, packB);
-        return run_gather();
-      }
-      if (rc_) return rc_;
-    }
-    Cg = tmpC;
   }
 
   bool mfma = mfma_shape;
@@ -2259,6 +2453,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
   HIP_CHECK(hipSetDevice(net->device));
   WsCtx ws{net->arena.base ? &net->arena : nullptr, net->stream, capture,
            false, deferred};
+  ws.stream2 = net->stream2;
   if (net->has_final && net->final_t.owned) {
     if (net->final_in_arena)
       net->arena.release(net->final_t.data);
@@ -2305,6 +2500,7 @@ static int contract_impl(tn_net* net, const u64* pairs, size_t nsteps,
   std::vector<Pending> pending;
   hipEvent_t pending_done = nullptr;
   std::vector<hipEvent_t> prep_events;
+  ws.events = &prep_events;  // pipeline events: destroyed after the walk
   const bool overlap =
       !prepack_disabled() && net->stream2 && net->arena.base != nullptr;
 

@@ -19,10 +19,10 @@ from .tensor import CompositeTensor, LeafTensor
 
 class StepInfo:
     __slots__ = ("i", "j", "m", "n", "k", "flops", "out_legs", "out_dims",
-                 "packa", "packb")
+                 "packa", "packb", "pipe_p")
 
     def __init__(self, i, j, m, n, k, flops, out_legs, out_dims,
-                 packa=False, packb=False):
+                 packa=False, packb=False, pipe_p=0):
         self.i, self.j = i, j
         self.m, self.n, self.k = m, n, k
         self.flops = flops
@@ -30,6 +30,39 @@ class StepInfo:
         self.out_dims = out_dims
         self.packa = packa  # TTGT would permute A into [M..][K..] order
         self.packb = packb  # TTGT would permute B into [K..][N..] order
+        self.pipe_p = pipe_p  # pack-pipeline window count (0 = serial pack)
+
+
+def _pipeline_windows(m, n, k, packa, packb, kdims, esize):
+    """Mirror of the C dispatch's pack-pipeline model (tnc_hip.hip
+    einsum_dev_impl): number of K windows the pipelined TTGT would use,
+    0 when the step packs serially. Keep in sync with the constants there
+    (permute ~4 TB/s, stream ~6 TB/s, window tiles >= 256)."""
+    if not packa:
+        return 0
+    m, n, k = int(m), int(n), int(k)
+    if not (m >= 32 and n >= 32) or m % 128 or n % 64 or k % 16:
+        return 0
+    gemm_worthy = k >= 16 and m >= 128 and n >= 64
+    if (k <= 64 or m < 16 or n < 16) and not gemm_worthy:
+        return 0
+    packbytes = m * k * esize + (k * n * esize if packb else 0)
+    tiles_w = (m // 128) * ((n + 63) // 64)
+    best_save, best_p = 0.0, 0
+    p = 1
+    for d in kdims:
+        if p >= 16:
+            break
+        p *= int(d)
+        if p < 2 or p > 16:
+            continue
+        if (k // p) % 16 or tiles_w < 256:
+            continue
+        save = (2.0 * packbytes / 4e12 * (1 - 1.0 / p)
+                - 2.0 * p * (m * n * esize) / 6e12)
+        if save > best_save:
+            best_save, best_p = save, p
+    return best_p if best_save > 1e-4 else 0
 
 
 def plan_steps(leaves, steps):
@@ -52,11 +85,14 @@ def plan_steps(leaves, steps):
         a_axes += [a.legs.index(l) for l in shared_a]
         b_axes = [b.legs.index(l) for l in shared_a]
         b_axes += [x for x, l in enumerate(b.legs) if l not in aset]
+        packa = a_axes != list(range(len(a.legs)))
+        packb = b_axes != list(range(len(b.legs)))
+        kdims = [a.bond_dims[a.legs.index(l)] for l in shared_a]
+        # window count is esize-invariant (the save model scales linearly)
+        pipe_p = _pipeline_windows(m, n, k, packa, packb, kdims, 16)
         infos.append(
             StepInfo(i, j, m, n, k, contract_cost_tensors(a, b), out.legs,
-                     out.bond_dims,
-                     packa=a_axes != list(range(len(a.legs))),
-                     packb=b_axes != list(range(len(b.legs))))
+                     out.bond_dims, packa=packa, packb=packb, pipe_p=pipe_p)
         )
         views[i] = out
         views[j] = None
@@ -83,7 +119,9 @@ def _step_ws_bytes(info, esize):
     MF_TN=64 in tnc_amd/csrc/tnc_hip.hip."""
     if info.m == 1 and info.n == 1 and info.k > 64:
         return 1 << 25  # dot partial buffer (<= 2^21 blocks * 16 B)
-    return _pack_ws_bytes(info, esize)
+    # pipelined steps additionally hold pipe_p split-K-style output slices
+    return (_pack_ws_bytes(info, esize)
+            + info.pipe_p * int(info.m) * int(info.n) * esize)
 
 
 def arena_bytes(leaves, steps, infos, esize=16):
