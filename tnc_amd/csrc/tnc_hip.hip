@@ -1716,7 +1716,15 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
         }
       }
     }
-    if (best_p && best_save > 1e-4) {
+    // measured gate (r02 A/B on hardware): marginal steps LOSE to the
+    // serial pack (window-launch overhead + slice-reduce traffic), so
+    // pipeline only when the predicted win is substantial and the pack
+    // dwarfs the output (rqc36's 4-10x-ratio steps regressed ~12 ms;
+    // syc49's 64x step gains ~12 ms)
+    if (best_p && (best_save <= 5e-3 ||
+                   (double)packbytes < 8.0 * (double)(M * N * sizeof(CT))))
+      best_p = 0;
+    if (best_p) {
       const int P = best_p;
       const u64 kc = best_kc;
       CT* Awin = nullptr;

@@ -62,7 +62,10 @@ def _pipeline_windows(m, n, k, packa, packb, kdims, esize):
                 - 2.0 * p * (m * n * esize) / 6e12)
         if save > best_save:
             best_save, best_p = save, p
-    return best_p if best_save > 1e-4 else 0
+    # measured gate (see tnc_hip.hip): only strongly-favorable steps win
+    if best_save <= 5e-3 or packbytes < 8.0 * m * n * esize:
+        return 0
+    return best_p
 
 
 def plan_steps(leaves, steps):
