@@ -1763,7 +1763,7 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       } else {
         // mixed-radix suffix products of the leading K legs (window digit
         // x has radix A.dims[k_a[x]])
-        u64 sufA[16] = {}, sufB[16] = {};
+        u64 sufA[16] = {};
         {
           u64 s = 1;
           for (int x = best_npre - 1; x >= 0; --x) {
