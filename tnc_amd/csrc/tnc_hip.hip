@@ -1877,7 +1877,10 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       int rc_ = ws_alloc(ws, (void**)&packA, elems * sizeof(CT));
       // pack workspace doesn't fit -> run the shape through the (slower)
       // gather kernels instead of failing the whole contraction
-      if (rc_ == TN_ERR_OOM && gather_ok) return run_gather();
+      if (rc_ == TN_ERR_OOM && gather_ok) {
+        ws_free(ws, tmpC);
+        return run_gather();
+      }
       if (rc_) return rc_;
     }
     {
@@ -1894,6 +1897,7 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
       int rc_ = ws_alloc(ws, (void**)&packB, elems * sizeof(CT));
       if (rc_ == TN_ERR_OOM && gather_ok) {
         ws_free(ws, packA);
+        ws_free(ws, tmpC);
         return run_gather();
       }
       if (rc_) return rc_;
