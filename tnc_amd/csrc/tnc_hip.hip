@@ -1345,6 +1345,18 @@ static bool pipeline_disabled() {
   return v == 1;
 }
 
+// testing hook: run every shape-feasible step through the pipeline,
+// ignoring the profitability gate (exercises the window/event machinery
+// on small cases the gate would reject)
+static bool pipeline_forced() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TN_PIPELINE_FORCE");
+    v = (e && e[0] && e[0] != '0') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 // The gather-staged GEMM is OPT-IN (TN_GATHER_GEMM=1): measured on the
 // rqc36 dominant shapes it is ~35% SLOWER than pack + pure GEMM (44.9 vs
 // 69.6 TF/s on M16384/N4096/K32768) because the GEMM re-reads its operand
@@ -1696,8 +1708,8 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     u64 packbytes =
         ((needA ? M * K : 0) + (needB ? K * N : 0)) * sizeof(CT);
     double best_save = 0.0;
-    int best_p = 0, best_npre = 0;
-    u64 best_kc = 0;
+    int best_p = 0, best_npre = 0, feas_p = 0, feas_npre = 0;
+    u64 best_kc = 0, feas_kc = 0;
     {
       u64 p = 1;
       u64 tiles_w = (M / MF_T) * ((N + MF_TN - 1) / MF_TN);
@@ -1708,6 +1720,9 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
         if (kc % MF_K || tiles_w < 256) continue;
         double save = 2.0 * (double)packbytes / 4e12 * (1.0 - 1.0 / p) -
                       2.0 * p * (double)(M * N * sizeof(CT)) / 6e12;
+        feas_p = (int)p;
+        feas_kc = kc;
+        feas_npre = (int)(x + 1);
         if (save > best_save) {
           best_save = save;
           best_p = (int)p;
@@ -1721,8 +1736,13 @@ static int einsum_dev_impl(const u64* out_labels, const u64* out_shape,
     // pipeline only when the predicted win is substantial and the pack
     // dwarfs the output (rqc36's 4-10x-ratio steps regressed ~12 ms;
     // syc49's 64x step gains ~12 ms)
-    if (best_p && (best_save <= 5e-3 ||
-                   (double)packbytes < 8.0 * (double)(M * N * sizeof(CT))))
+    if (pipeline_forced() && feas_p) {
+      best_p = feas_p;
+      best_kc = feas_kc;
+      best_npre = feas_npre;
+    } else if (best_p &&
+               (best_save <= 5e-3 ||
+                (double)packbytes < 8.0 * (double)(M * N * sizeof(CT))))
       best_p = 0;
     if (best_p) {
       const int P = best_p;
