@@ -319,6 +319,65 @@ def run_single(args, fixture=None, steps=None, warmup=None, secondary=False):
     emit(line)
 
 
+def _fanin_secondary(args, tn, frozen_path, meta, rank, world, dev_id,
+                     device, backend, dist_t, torch, useful_flops, dtype):
+    """Partition fan-in mechanism (tree-cut plan, one partition per rank,
+    P2P exchange of open-leg intermediates, final on rank 0). The timed
+    region includes per-iteration engine setup (leaf scatter) like the
+    reference's in-run scatter; tree-cut leaves the heavy top-of-tree
+    merges sequential, so this line documents WHY slicing is the headline
+    mechanism (DESIGN.md)."""
+    from tnc_amd.dist import make_tree_plan
+    from tnc_amd.dist_gpu import run_fanin_gpu
+
+    plan = make_tree_plan(tn, frozen_path, world)
+    steps = max(1, min(args.steps, 3))
+    warmup = 1
+    times = []
+    for it in range(warmup + steps):
+        dist_t.barrier()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        handle = run_fanin_gpu(plan, rank, world, dist_t, torch, device,
+                               dev_id=dev_id, dtype=dtype, backend=backend)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        del handle
+        dt_t = torch.tensor([dt], dtype=torch.float64,
+                            device=device if backend == "nccl" else "cpu")
+        dist_t.all_reduce(dt_t, op=dist_t.ReduceOp.MAX)
+        if it >= warmup:
+            times.append(dt_t.item())
+    if rank == 0:
+        wall = sum(times)
+        emit({
+            "metric": f"pairwise-contraction GFLOP/s ({dtype})",
+            "value": useful_flops * steps / wall / 1e9,
+            "unit": "GFLOP/s",
+            "n_gpus": world,
+            "steps": steps,
+            "warmup": warmup,
+            "ms_per_step": wall / steps * 1e3,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "workload": f"{args.fixture}: partition fan-in mechanism "
+                            "(communication.rs:199-249 semantics; tree-cut "
+                            f"{plan.nparts}-way plan, per-iteration scatter "
+                            "included in the timed region)",
+                "mechanism": "partition-fanin",
+                "partitions": plan.nparts,
+                "metric_flops_per_contraction": useful_flops,
+            },
+            "roofline": None,
+            "cpu_baseline": None,
+            "secondary": True,
+        })
+
+
 def run_distributed(args):
     """N ranks, one GPU each (torch.distributed / RCCL). Parallelism is
     EDGE SLICING (tnc_amd/slicing.py): ceil(log2(N)) shared edges of the
@@ -364,6 +423,19 @@ def run_distributed(args):
     leaves, steps, _ = flatten_network(tn, replace)
     plan_infos = plan_steps(leaves, steps)
     useful_flops = sum(i.flops for i in plan_infos)
+
+    # SECONDARY mechanism line: partition fan-in (the reference's MPI
+    # shape, communication.rs:199-249) measured beside the slicing
+    # headline so an 8-GPU run records both. Fail-safe: any error here
+    # only drops this line, never the headline.
+    if not os.environ.get("TN_NO_FANIN"):
+        try:
+            _fanin_secondary(args, tn, frozen_path, meta, rank, world,
+                             dev_id, device, backend, dist_t, torch,
+                             useful_flops, dtype)
+        except Exception as e:  # noqa: BLE001
+            print(f"[bench] fan-in secondary failed: {e!r}",
+                  file=sys.stderr, flush=True)
     # final-tensor element count from the PLAN metadata (identical on every
     # rank, sliced or not — sliced edges are internal, the final view is the
     # network's): ranks whose assignment share is empty still allocate the

@@ -133,6 +133,9 @@ int tn_net_contract_profiled(tn_net* net, const uint64_t* pairs, size_t nsteps,
  * buffers with an external allocator, e.g. RCCL-communicated tensors). */
 int tn_memcpy_dtod(void* dst, const void* src, uint64_t bytes);
 
+/* Synchronous device-to-host copy (fetch an exchanged intermediate). */
+int tn_memcpy_dtoh(void* host_dst, const void* dev_src, uint64_t bytes);
+
 /* Metadata of the final tensor after the last contract call. labels/dims
  * must have room for 64 entries. */
 int tn_net_result_meta(tn_net* net, uint64_t* labels, uint64_t* dims,

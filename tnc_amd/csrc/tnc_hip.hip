@@ -2919,6 +2919,12 @@ extern "C" int tn_memcpy_dtod(void* dst, const void* src, u64 bytes) {
   return TN_OK;
 }
 
+extern "C" int tn_memcpy_dtoh(void* host_dst, const void* dev_src,
+                              u64 bytes) {
+  HIP_CHECK(hipMemcpy(host_dst, dev_src, bytes, hipMemcpyDeviceToHost));
+  return TN_OK;
+}
+
 extern "C" int tn_net_result_meta(tn_net* net, u64* labels, u64* dims,
                                   size_t* ndim) {
   if (!net || !net->has_final) FAILV(TN_ERR_INVALID, "no result available");

@@ -77,6 +77,10 @@ def lib() -> ctypes.CDLL:
         L.tn_memcpy_dtod.argtypes = [
             ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
         ]
+        L.tn_memcpy_dtoh.restype = ctypes.c_int
+        L.tn_memcpy_dtoh.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
+        ]
         L.tn_net_result_meta.restype = ctypes.c_int
         L.tn_net_result_meta.argtypes = [
             ctypes.c_void_p, u64p, u64p, ctypes.POINTER(ctypes.c_size_t),
