@@ -169,6 +169,12 @@ class ContractionEngine:
         if not self.net:
             raise RuntimeError(f"tn_net_create failed: {hiplib.last_error()}")
         reserve = arena_bytes(leaves, steps, self.infos, self.esize)
+        # launch-bound walks (many tiny steps) only replay as a hipGraph
+        # when every workspace comes from the arena — reserve a floor so
+        # replay engages (rqc24: ~2.4 ms/contraction unreserved vs
+        # ~0.6 ms replayed)
+        if len(steps) >= 64:
+            reserve = max(reserve, 256 * 1024 * 1024)
         # arena is an optimization: if the reservation fails (tiny GPUs,
         # fragmented memory), every block falls back to plain hipMalloc
         # (the stream-ordered pool is avoided — see ws_alloc in tnc_hip.hip)
