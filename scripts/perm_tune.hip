@@ -34,6 +34,8 @@ struct PermPerm {
   u64 aD[8];
 };
 
+__device__ unsigned long long g_bad[4];
+
 template <int AB, int BB, int THREADS>
 __global__ __launch_bounds__(THREADS) void k_tile(
     const double2* __restrict__ src, double2* __restrict__ dst, PermPerm pp) {
@@ -69,18 +71,33 @@ __global__ __launch_bounds__(THREADS) void k_tile(
     }
   }
   __syncthreads();
+  const u64 LIM = 1ull << 29;
   for (int e = tid; e < AN * BN; e += THREADS) {
     const int a = e & (AN - 1), b = e >> AB;
-    tile[b * AN + ((a ^ b) & (AN - 1))] = src[baseS + sboffS[b] + (u64)a];
+    u64 idx = baseS + sboffS[b] + (u64)a;
+    if (idx >= LIM) {
+      g_bad[0] = 1;
+      g_bad[1] = idx;
+      continue;
+    }
+    tile[b * AN + ((a ^ b) & (AN - 1))] = src[idx];
   }
   __syncthreads();
   for (int e = tid; e < AN * BN; e += THREADS) {
     const int b = e & (BN - 1), a = e >> BB;
-    dst[baseD + saoffD[a] + sboffD[b]] = tile[b * AN + ((a ^ b) & (AN - 1))];
+    u64 idx = baseD + saoffD[a] + sboffD[b];
+    if (idx >= LIM) {
+      g_bad[2] = 1;
+      g_bad[3] = idx;
+      continue;
+    }
+    dst[idx] = tile[b * AN + ((a ^ b) & (AN - 1))];
   }
 }
 
-int main() {
+int main(int argc, char** argv) {
+  const int only = argc > 1 ? atoi(argv[1]) : -1;
+  int vidx = 0;
   const int NB = 29;  // 2^29 c128 = 8.6 GB
   const u64 elems = 1ull << NB;
   double2 *src, *dst;
@@ -91,15 +108,19 @@ int main() {
   // permutation: dst bit i <- src bit perm[i]; model an interleaved pack
   // (like a_axes reordering): src bits shuffled with stride-2 interleave
   int perm[NB];
-  {
-    int half = NB / 2, k = 0;
-    for (int i = 0; i < half; ++i) {
-      perm[k++] = i * 2;  // even src bits first
-    }
-    for (int i = 0; i < NB - half; ++i) perm[k++] = i * 2 + 1;
+  const int shape = argc > 2 ? atoi(argv[2]) : 0;
+  if (shape == 0) {  // interleave (rqc36 step-437-like)
+    int k = 0;
+    for (int i = 0; i * 2 < NB; ++i) perm[k++] = i * 2;
+    for (int i = 0; i * 2 + 1 < NB; ++i) perm[k++] = i * 2 + 1;
+  } else {  // bit reversal (worst case)
+    for (int i = 0; i < NB; ++i) perm[i] = NB - 1 - i;
   }
   // srcStrideOfDstBit[i] = 1 << perm[i]
   auto run = [&](int AB, int BB, int THREADS, auto kern) {
+    if (only >= 0 && vidx++ != only) return;
+    printf("running AB=%d BB=%d THREADS=%d\n", AB, BB, THREADS);
+    fflush(stdout);
     PermPerm pp{};
     // a-bits: dst-low? a-bits must be the SRC-low AB bits (coalesced src):
     // find dst bits whose src stride is 1..1<<(AB-1)
@@ -138,6 +159,11 @@ int main() {
     CHECK(hipDeviceSynchronize());
     float ms = 0;
     CHECK(hipEventElapsedTime(&ms, e0, e1));
+    u64 bad[4] = {};
+    CHECK(hipMemcpyFromSymbol(bad, HIP_SYMBOL(g_bad), sizeof bad));
+    if (bad[0] || bad[2])
+      printf("  BAD: src_oob=%llu (idx %llu) dst_oob=%llu (idx %llu)\n",
+             bad[0], bad[1], bad[2], bad[3]);
     double gbs = 5.0 * 2.0 * elems * 16 / (ms / 1e3) / 1e9;
     printf("AB=%d BB=%d THREADS=%4d  %7.1f GB/s (%.2f ms/pass)\n", AB, BB,
            THREADS, gbs, ms / 5.0);
@@ -153,5 +179,9 @@ int main() {
   run(5, 5, 512, k_tile<5, 5, 512>);
   run(6, 7, 512, k_tile<6, 7, 512>);  // 128 KB LDS: 1 block/CU
   run(7, 6, 512, k_tile<7, 6, 512>);
+  run(6, 5, 1024, k_tile<6, 5, 1024>);
+  run(5, 6, 1024, k_tile<5, 6, 1024>);
+  run(5, 5, 1024, k_tile<5, 5, 1024>);
+  run(6, 4, 512, k_tile<6, 4, 512>);
   return 0;
 }

@@ -212,7 +212,10 @@ __global__ void k_einsum_anyk(const CT* __restrict__ A,
 // index of a coalesced WRITE, staged through a 64x64 XOR-swizzled LDS tile
 // (64 KB c128 -> 2 blocks/CU).
 #define TN_PERM_AB 6
-#define TN_PERM_BB 6
+// BB=5 (64x32 tiles, 32 KB LDS -> more blocks/CU) measured 5.2 TB/s vs
+// 4.3 TB/s at 6/6 on the rqc36-interleave shape and holds 5.2 on bit
+// reversal (scripts/perm_tune.hip scan, r02)
+#define TN_PERM_BB 5
 #define TN_PERM_MAXBITS 34
 
 struct PermPerm {
@@ -227,10 +230,11 @@ struct PermPerm {
 template <typename CT>
 __global__ __launch_bounds__(512) void k_permute_tile(
     const CT* __restrict__ src, CT* __restrict__ dst, PermPerm pp) {
-  __shared__ CT tile[64 * 64];
-  __shared__ u64 sboffS[64], sboffD[64], saoffD[64];
+  constexpr int AN = 1 << TN_PERM_AB, BN = 1 << TN_PERM_BB;
+  __shared__ CT tile[AN * BN];
+  __shared__ u64 sboffS[BN], sboffD[BN], saoffD[AN];
   const int tid = threadIdx.x;
-  if (tid < 64) {
+  if (tid < BN) {
     u64 os = 0, od = 0;
     for (int i = 0; i < TN_PERM_BB; ++i)
       if (tid >> i & 1) {
@@ -239,8 +243,8 @@ __global__ __launch_bounds__(512) void k_permute_tile(
       }
     sboffS[tid] = os;
     sboffD[tid] = od;
-  } else if (tid < 128) {
-    const int a = tid - 64;
+  } else if (tid < BN + AN) {
+    const int a = tid - BN;
     u64 od = 0;
     for (int i = 0; i < TN_PERM_AB; ++i)
       if (a >> i & 1) od += pp.aD[i];
@@ -258,14 +262,16 @@ __global__ __launch_bounds__(512) void k_permute_tile(
     }
   }
   __syncthreads();
-  for (int e = tid; e < 64 * 64; e += 512) {
-    const int a = e & 63, b = e >> 6;  // lane ~ a: coalesced src reads
-    tile[b * 64 + (a ^ b)] = src[baseS + sboffS[b] + (u64)a];
+  for (int e = tid; e < AN * BN; e += 512) {
+    // lane ~ a: coalesced src reads
+    const int a = e & (AN - 1), b = e >> TN_PERM_AB;
+    tile[b * AN + ((a ^ b) & (AN - 1))] = src[baseS + sboffS[b] + (u64)a];
   }
   __syncthreads();
-  for (int e = tid; e < 64 * 64; e += 512) {
-    const int b = e & 63, a = e >> 6;  // lane ~ b: coalesced dst writes
-    dst[baseD + saoffD[a] + sboffD[b]] = tile[b * 64 + (a ^ b)];
+  for (int e = tid; e < AN * BN; e += 512) {
+    // lane ~ b: coalesced dst writes
+    const int b = e & (BN - 1), a = e >> TN_PERM_BB;
+    dst[baseD + saoffD[a] + sboffD[b]] = tile[b * AN + ((a ^ b) & (AN - 1))];
   }
 }
 
