@@ -271,3 +271,38 @@ def test_peps_gpu_vs_oracle():
     ref = contract_network(network_to_otensors(tn), replace)
     assert legs == [] == ref.legs
     np.testing.assert_allclose(data, ref.data, rtol=1e-10)
+
+
+def test_engine_fuzz_random_circuits():
+    """Engine-level fuzz: random circuits of varying density/seed through
+    the full device walk (prepack overlap, TTGT routing, graph arming)
+    against the oracle. Broader coverage for the cross-step machinery
+    than the fixed fixtures."""
+    from tnc_amd import RandomGreedy
+    from tnc_amd.builders import random_circuit
+    from tnc_amd.connectivity import ConnectivityLayout
+    from tnc_amd.executor import ContractionEngine
+
+    cases = [
+        (13, 9, 0.4, 0.7, 101),
+        (14, 7, 0.6, 0.5, 102),
+        (12, 11, 0.5, 0.8, 103),
+        (15, 8, 0.3, 0.6, 104),
+        (11, 12, 0.7, 0.7, 105),
+        (16, 6, 0.5, 0.4, 106),
+    ]
+    for q, r, p1, p2, seed in cases:
+        tn = random_circuit(q, r, p1, p2, seed, ConnectivityLayout.EAGLE)
+        replace = RandomGreedy(4, seed=seed).find_path(tn).replace_path()
+        ref = contract_network(network_to_otensors(tn), replace)
+        eng = ContractionEngine(tn, replace)
+        try:
+            eng.contract()
+            _, first = eng.result()
+            eng.contract()  # second pass: graph capture+replay path
+            _, second = eng.result()
+        finally:
+            eng.close()
+        np.testing.assert_allclose(first, ref.data, rtol=1e-10, atol=1e-12,
+                                   err_msg=f"case {(q, r, p1, p2, seed)}")
+        np.testing.assert_array_equal(first, second)
