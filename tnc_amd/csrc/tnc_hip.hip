@@ -1289,9 +1289,14 @@ static bool permute_tiled(const CT* src, CT* dst, u64 elems,
             [&](int x, int y) { return bits[x].ss < bits[y].ss; });
   std::sort(byD.begin(), byD.end(),
             [&](int x, int y) { return bits[x].sd < bits[y].sd; });
-  for (int i = 0; i < nb; ++i)
-    if (bits[byS[i]].ss != (1ull << i) || bits[byD[i]].sd != (1ull << i))
-      return false;  // not a contiguous span on both sides
+  // the a-bits (lane index) must be the TN_PERM_AB lowest SOURCE strides,
+  // exactly 1..2^(AB-1), for coalesced reads. All other bits may carry
+  // ARBITRARY source strides (the kernel sums per-bit offsets), which
+  // admits sub-box permutes — e.g. the pack pipeline's K-windows, whose
+  // fixed leading legs leave gaps in the source span. The destination
+  // side is a full span by construction (suffix products above).
+  for (int i = 0; i < TN_PERM_AB; ++i)
+    if (bits[byS[i]].ss != (1ull << i)) return false;
   PermPerm pp;
   std::vector<char> used(nb, 0);
   for (int i = 0; i < TN_PERM_AB; ++i) {
